@@ -1,0 +1,66 @@
+"""Multi-process test helper: run a function on N gloo ranks on CPU.
+
+Used by the FSDP parity tests so the distributed control flow (the real
+all-gather / reduce-scatter paths) is exercised without GPUs, matching
+BASELINE.json config 1 (plumbing on CPU/gloo).
+"""
+
+import multiprocessing as mp
+import os
+import pickle
+import tempfile
+import traceback
+
+
+def _worker(rank, world_size, port, fn, args, result_dir):
+    try:
+        os.environ["RANK"] = str(rank)
+        os.environ["LOCAL_RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world_size)
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        result = fn(rank, world_size, *args)
+        with open(os.path.join(result_dir, f"rank{rank}.pkl"), "wb") as f:
+            pickle.dump(("ok", result), f)
+    except Exception:
+        with open(os.path.join(result_dir, f"rank{rank}.pkl"), "wb") as f:
+            pickle.dump(("err", traceback.format_exc()), f)
+        raise
+
+
+_PORT = [29600]
+
+
+def run_multiprocess(fn, world_size=2, args=(), timeout=300):
+    """Run fn(rank, world_size, *args) on `world_size` spawned processes.
+
+    Returns the list of per-rank return values (picklable)."""
+    ctx = mp.get_context("spawn")
+    _PORT[0] += 1
+    port = _PORT[0] + (os.getpid() % 500)
+    with tempfile.TemporaryDirectory() as result_dir:
+        procs = [
+            ctx.Process(
+                target=_worker,
+                args=(r, world_size, port, fn, args, result_dir),
+            )
+            for r in range(world_size)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout)
+        for r, p in enumerate(procs):
+            if p.is_alive():
+                p.terminate()
+                raise TimeoutError(f"rank {r} timed out")
+        results = []
+        for r in range(world_size):
+            path = os.path.join(result_dir, f"rank{r}.pkl")
+            assert os.path.exists(path), f"rank {r} produced no result (exit {procs[r].exitcode})"
+            with open(path, "rb") as f:
+                status, payload = pickle.load(f)
+            if status == "err":
+                raise RuntimeError(f"rank {r} failed:\n{payload}")
+            results.append(payload)
+        return results

@@ -1,0 +1,21 @@
+from ._extension import ext, has_ext, use_hip
+from .layernorm import LayerNorm, layer_norm
+from .attention import attention, math_attention
+from .cross_entropy import CrossEntropyLoss, cross_entropy
+from .adamw import FusedAdamW
+from .multi_tensor import local_sqnorm, scale_
+
+__all__ = [
+    "ext",
+    "has_ext",
+    "use_hip",
+    "LayerNorm",
+    "layer_norm",
+    "attention",
+    "math_attention",
+    "CrossEntropyLoss",
+    "cross_entropy",
+    "FusedAdamW",
+    "local_sqnorm",
+    "scale_",
+]

@@ -1,0 +1,68 @@
+"""Multi-head attention core (kernel K4 in SURVEY.md §2D).
+
+The reference gets its attention from the timm Block: per-head
+softmax(Q K^T * d^-0.5) V with optional attention dropout
+(run_vit_training.py:134-141).  For the 10B config head_dim = 5120/32 =
+160 — off the usual 64/128 flash fast paths — so the GPU path is our
+own CDNA4 flash-style kernel (MFMA 16x16x32 bf16 tiles, LDS-staged K/V,
+online softmax, O(T) memory; backward recomputes the forward tiles).
+
+No Triton, no aotriton SDPA: the CPU/no-ext fallback is an explicit
+math composition (matmul + softmax), which is also the numerics
+reference for the kernel tests.
+"""
+
+import torch
+
+from ._extension import ext, use_hip
+
+
+def math_attention(q, k, v, scale=None, dropout_p=0.0, training=False):
+    """Explicit-math reference path: q,k,v [B, H, T, D]."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    scores = torch.matmul(q, k.transpose(-2, -1)) * scale
+    # softmax in fp32 for bf16 inputs (matches the kernel's fp32 online
+    # softmax accumulation)
+    probs = torch.softmax(scores.float(), dim=-1).to(q.dtype)
+    if dropout_p > 0.0 and training:
+        probs = torch.nn.functional.dropout(probs, p=dropout_p)
+    return torch.matmul(probs, v)
+
+
+class _FlashAttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        o, lse = ext().fmha_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = ext().fmha_bwd(do.contiguous(), q, k, v, o, lse, ctx.scale)
+        return dq, dk, dv, None
+
+
+def attention(q, k, v, scale=None, dropout_p=0.0, training=False):
+    """Attention core on [B, H, T, D] tensors.
+
+    GPU: our flash-style HIP kernel (dropout_p must be 0 there for now —
+    the 10B recipe uses att_dropout 0.0; nonzero dropout falls back to
+    the math path with a warning-free explicit mask).
+    CPU: math composition.
+    """
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    use_kernel = (
+        dropout_p == 0.0
+        and q.dtype in (torch.bfloat16, torch.float16)
+        and use_hip(q, k, v)
+        and hasattr(ext(), "fmha_fwd")
+    )
+    if use_kernel:
+        return _FlashAttentionFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), scale
+        )
+    return math_attention(q, k, v, scale, dropout_p, training)

@@ -1,0 +1,52 @@
+"""LayerNorm over the last dimension (kernel K2 in SURVEY.md §2D).
+
+Replaces the torch/timm LayerNorm the reference uses inside every Block
+and for the final norm (reference run_vit_training.py:151, timm Block
+norm1/norm2).  On GPU the forward/backward run in our CDNA4 HIP kernels
+(one-pass mean/var with fp32 accumulation, vectorized bf16x8 loads); on
+CPU we defer to torch's native op, which keeps autograd semantics
+identical for the no-GPU test suite.
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ._extension import ext, use_hip
+
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        y, mean, rstd = ext().layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ext().layernorm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+def layer_norm(x, weight, bias, eps=1e-6):
+    if use_hip(x):
+        return _LayerNormFn.apply(x.contiguous(), weight, bias, eps)
+    return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+class LayerNorm(nn.Module):
+    """Drop-in LayerNorm module backed by the HIP kernel on GPU."""
+
+    def __init__(self, dim, eps=1e-6):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+        self.dim = dim
+
+    def forward(self, x):
+        return layer_norm(x, self.weight, self.bias, self.eps)
+
+    def extra_repr(self):
+        return f"{self.dim}, eps={self.eps}"

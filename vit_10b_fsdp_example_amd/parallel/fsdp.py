@@ -1,0 +1,436 @@
+"""From-scratch ZeRO-3 FSDP engine for MI355X.
+
+Capability parity target: torch_xla's XlaFullyShardedDataParallel as the
+reference uses it (run_vit_training.py:13,177-181; utils.py:8,29) —
+per-module parameter sharding, all-gather before forward and backward,
+reshard after forward, reduce-scatter of gradients, full-norm gradient
+clipping, shard metadata for offline consolidation, CPU-side wrapping.
+
+The design is MI355X-first rather than a translation:
+
+  * One flat buffer per wrapped unit.  Every parameter of a unit lives
+    in ONE flat fp32 master shard (1/world_size each rank, padded), and
+    materializes into ONE flat compute-dtype (bf16) buffer per unit.
+    Collectives therefore move one ~630 MB message per ViT-10B block —
+    the large-message regime where RCCL's multi-link xGMI algorithms
+    reach aggregate link bandwidth — instead of 13 per-tensor calls.
+  * Mixed precision natively: fp32 master shards + AdamW state, bf16
+    gather/compute/reduce-scatter.  compute_dtype=fp32 gives the exact
+    reference numerics (used by the CPU parity tests).
+  * Overlap by construction: param gathers for the NEXT unit are issued
+    asynchronously (prefetch) while the current unit computes, in both
+    forward and backward, on a dedicated RCCL communicator; gradient
+    reduce-scatters go on a second communicator and are waited only at
+    end-of-backward finalization.
+  * Storage lifecycle, not graph tricks: the unit's full flat buffer is
+    a single allocation whose storage is freed (resize_(0)) on reshard
+    and refilled in the pre-backward hook; module params are stable
+    leaf *views* into it, so gradient-checkpoint recompute reads the
+    refilled weights with zero extra bookkeeping.
+
+Gradient flow: the leaf views accumulate .grad normally; a
+post-accumulate-grad hook copies each one into the unit's flat grad
+buffer and, when the unit is complete, frees the full params and
+launches the async reduce-scatter.  A backward-engine callback
+(registered by the root's pre-backward hook) waits for all pending
+reductions and installs fp32 mean gradients on the master shards, so
+`optimizer.step()` needs no knowledge of any of this.
+"""
+
+import torch
+import torch.nn as nn
+
+from ..ops import local_sqnorm, scale_
+from .comm import CommContext
+
+# Modules currently executing an FSDP forward (root detection).
+_EXEC_STACK = []
+
+
+def _free_storage(t):
+    if t.untyped_storage().size() > 0:
+        t.untyped_storage().resize_(0)
+
+
+def _alloc_storage(t, nbytes):
+    if t.untyped_storage().size() == 0:
+        t.untyped_storage().resize_(nbytes)
+
+
+class FullyShardedDataParallel(nn.Module):
+    WRAPPER_ATTR = "_fsdp_wrapped_module"
+
+    def __init__(
+        self,
+        module,
+        reshard_after_forward=True,
+        flatten_parameters=True,
+        compute_dtype=torch.float32,
+        device=None,
+        prefetch=True,
+    ):
+        super().__init__()
+        # `flatten_parameters` is accepted for CLI compatibility with the
+        # reference (run_vit_training.py:180,359).  This engine is
+        # flat-by-design: a unit's parameters always live in one flat
+        # buffer, because one large RCCL message per unit is the xGMI-
+        # efficient shape.  The flag therefore has no effect (documented
+        # deviation; consolidation understands our layout either way).
+        self.reshard_after_forward = reshard_after_forward
+        self.flatten_parameters = flatten_parameters
+        self.compute_dtype = compute_dtype
+        self.prefetch = prefetch
+
+        self._comm = CommContext.get()
+        ws, rank = self._comm.world_size, self._comm.rank
+
+        if device is not None:
+            self.device = torch.device(device)
+        else:
+            p = next(module.parameters(), None)
+            self.device = p.device if p is not None else torch.device("cpu")
+
+        if isinstance(module, FullyShardedDataParallel):
+            raise ValueError("do not wrap an FSDP instance directly in another FSDP")
+        setattr(self, self.WRAPPER_ATTR, module)
+
+        # ---- collect parameters owned by this unit (nested FSDP units
+        # manage their own and are pruned from the walk) ----
+        infos = []  # [submodule, name, shape, numel, offset, param]
+        offset = 0
+
+        def _collect(mod):
+            nonlocal offset
+            if isinstance(mod, FullyShardedDataParallel):
+                return
+            for name, p in list(mod.named_parameters(recurse=False)):
+                infos.append([mod, name, p.shape, p.numel(), offset, p])
+                offset += p.numel()
+            for child in mod.children():
+                _collect(child)
+
+        _collect(module)
+        total = offset
+        assert total > 0, "FSDP unit has no parameters to shard"
+        self._total_numel = total
+        self._padded_numel = -(-total // ws) * ws
+        self._shard_numel = self._padded_numel // ws
+
+        # ---- fp32 master shard, built on the module's current device so
+        # --shard_on_cpu never materializes full params on the GPU ----
+        src_device = infos[0][5].device
+        flat = torch.zeros(self._padded_numel, dtype=torch.float32, device=src_device)
+        for _, _, _, numel, off, p in infos:
+            flat.narrow(0, off, numel).copy_(p.detach().reshape(-1).to(torch.float32))
+        shard = (
+            flat.narrow(0, rank * self._shard_numel, self._shard_numel)
+            .to(self.device)
+            .clone()
+        )
+        del flat
+        self.flat_param = nn.Parameter(shard)
+
+        # ---- full compute-dtype buffer + stable leaf views ----
+        self._full_flat = torch.empty(
+            self._padded_numel, dtype=compute_dtype, device=self.device
+        )
+        # Collective writes go through a separate alias tensor with its
+        # own autograd version counter, so refilling the buffer in the
+        # pre-backward hook never trips saved-tensor version checks on
+        # the views that forward ops saved.
+        self._write_alias = torch.empty(0, dtype=compute_dtype, device=self.device)
+        self._write_alias.set_(
+            self._full_flat.untyped_storage(), 0, (self._padded_numel,)
+        )
+        self._elem_bytes = self._full_flat.element_size()
+
+        self._param_infos = []  # (submodule, name, shape, numel, offset)
+        self._views = []
+        with torch.no_grad():
+            for mod, name, shape, numel, off, _p in infos:
+                view = self._full_flat.narrow(0, off, numel).view(shape).detach()
+                view.requires_grad_(True)
+                del mod._parameters[name]
+                setattr(mod, name, view)
+                self._param_infos.append((mod, name, shape, numel, off))
+                self._views.append(view)
+        _free_storage(self._full_flat)
+
+        for idx, view in enumerate(self._views):
+            view.register_post_accumulate_grad_hook(self._make_grad_hook(idx))
+
+        # flat grad buffer (compute dtype); storage freed between backwards
+        self._full_grad = torch.empty(
+            self._padded_numel, dtype=compute_dtype, device=self.device
+        )
+        _free_storage(self._full_grad)
+
+        # runtime state
+        self._is_root = None
+        self._root_ref = None
+        self._fwd_order = None  # root only: units in forward execution order
+        self._pre_bwd_done = False
+        self._grads_arrived = 0
+        self._fresh = False  # full params match current master shard
+        self._pending_gather = None  # (work, src_ref)
+        self._pending_reduce = None  # (work, out_shard)
+
+    # ------------------------------------------------------------------
+    # materialization
+    # ------------------------------------------------------------------
+
+    def _comm_shard(self):
+        """The shard in comm/compute dtype (cast from the fp32 master)."""
+        if self.flat_param.dtype == self.compute_dtype:
+            return self.flat_param.data
+        return self.flat_param.data.to(self.compute_dtype)
+
+    def _resident(self):
+        return self._full_flat.untyped_storage().size() > 0
+
+    def _issue_gather(self, async_op):
+        _alloc_storage(self._full_flat, self._padded_numel * self._elem_bytes)
+        src = self._comm_shard()
+        work = self._comm.all_gather_into(self._write_alias, src, async_op=async_op)
+        return (work, src)
+
+    def _materialize(self):
+        """Ensure full params are resident and up to date."""
+        if self._pending_gather is not None:
+            work, _src = self._pending_gather
+            self._pending_gather = None
+            work.wait()
+            self._fresh = True
+            return
+        if self._resident() and self._fresh:
+            return
+        work, _src = self._issue_gather(async_op=False)
+        work.wait()
+        self._fresh = True
+
+    def _prefetch_gather(self):
+        """Issue this unit's param gather asynchronously (called by the
+        previous unit in execution order while it computes)."""
+        if self._pending_gather is not None or (self._resident() and self._fresh):
+            return
+        self._pending_gather = self._issue_gather(async_op=True)
+
+    def _free_full(self):
+        _free_storage(self._full_flat)
+
+    # ------------------------------------------------------------------
+    # forward
+    # ------------------------------------------------------------------
+
+    def forward(self, *args, **kwargs):
+        if self._is_root is None:
+            self._is_root = len(_EXEC_STACK) == 0
+            if self._is_root:
+                self._fwd_order = []
+        root = _EXEC_STACK[0] if _EXEC_STACK else self
+        self._root_ref = root
+        if root._fwd_order is not None and self not in root._fwd_order:
+            root._fwd_order.append(self)
+
+        _EXEC_STACK.append(self)
+        try:
+            self._pre_bwd_done = False
+            self._grads_arrived = 0
+            self._materialize()
+            self._prefetch_neighbor(root, direction=+1)
+            out = getattr(self, self.WRAPPER_ATTR)(*args, **kwargs)
+            grad_flows = self._register_pre_backward(out)
+            if self.reshard_after_forward:
+                self._free_full()
+            elif not grad_flows:
+                # no backward will refresh these params; drop them so a
+                # later forward re-gathers post-optimizer values
+                self._free_full()
+                self._fresh = False
+        finally:
+            _EXEC_STACK.pop()
+        return out
+
+    def _prefetch_neighbor(self, root, direction):
+        if not self.prefetch or root._fwd_order is None:
+            return
+        try:
+            i = root._fwd_order.index(self)
+        except ValueError:
+            return
+        j = i + direction
+        if 0 <= j < len(root._fwd_order):
+            root._fwd_order[j]._prefetch_gather()
+
+    def _register_pre_backward(self, out):
+        tensors = out if isinstance(out, (tuple, list)) else (out,)
+        for t in tensors:
+            if torch.is_tensor(t) and t.requires_grad:
+                t.register_hook(self._pre_backward_hook)
+                return True
+        return False
+
+    # ------------------------------------------------------------------
+    # backward
+    # ------------------------------------------------------------------
+
+    def _pre_backward_hook(self, grad):
+        if self._pre_bwd_done:
+            return grad
+        self._pre_bwd_done = True
+        if self._is_root:
+            torch.autograd.Variable._execution_engine.queue_callback(
+                self._finalize_backward_all
+            )
+        self._materialize()
+        _alloc_storage(self._full_grad, self._padded_numel * self._elem_bytes)
+        self._full_grad.zero_()
+        # prefetch the unit the backward will need next (previous in
+        # forward order)
+        if self._root_ref is not None:
+            self._prefetch_neighbor(self._root_ref, direction=-1)
+        return grad
+
+    def _make_grad_hook(self, idx):
+        def hook(view):
+            self._on_param_grad(idx, view)
+
+        return hook
+
+    def _on_param_grad(self, idx, view):
+        if view.grad is None:
+            return
+        _mod, _name, _shape, numel, off = self._param_infos[idx]
+        _alloc_storage(self._full_grad, self._padded_numel * self._elem_bytes)
+        self._full_grad.narrow(0, off, numel).copy_(view.grad.reshape(-1))
+        view.grad = None
+        self._grads_arrived += 1
+        if self._grads_arrived == len(self._views):
+            self._post_backward()
+
+    def _post_backward(self):
+        """All of this unit's param grads are in the flat buffer: free the
+        full params and launch the async reduce-scatter."""
+        self._free_full()
+        self._fresh = False  # master shard will change at optimizer.step
+        out_shard = torch.empty(
+            self._shard_numel, dtype=self.compute_dtype, device=self.device
+        )
+        work = self._comm.reduce_scatter_into(out_shard, self._full_grad, async_op=True)
+        self._pending_reduce = (work, out_shard)
+
+    def _finalize_unit(self):
+        """Wait the pending reduction, install the fp32 mean grad on the
+        master shard, release buffers."""
+        if self._grads_arrived and self._pending_reduce is None:
+            # partial grads (frozen subgraph): reduce what we have —
+            # missing slices are zeros from the buffer memset.
+            self._post_backward()
+        if self._pending_reduce is not None:
+            work, out_shard = self._pending_reduce
+            self._pending_reduce = None
+            work.wait()
+            g = out_shard.to(torch.float32)
+            if self._comm.world_size > 1:
+                g.div_(self._comm.world_size)
+            if self.flat_param.grad is None:
+                self.flat_param.grad = g
+            else:
+                self.flat_param.grad.add_(g)
+        _free_storage(self._full_grad)
+        self._grads_arrived = 0
+
+    def _finalize_backward_all(self):
+        for unit in self._all_units():
+            unit._finalize_unit()
+
+    def _all_units(self):
+        return [m for m in self.modules() if isinstance(m, FullyShardedDataParallel)]
+
+    # ------------------------------------------------------------------
+    # grad clipping (reference run_vit_training.py:270 — clip on the FULL
+    # gradient norm: local shard sq-norm in one multi-tensor kernel, one
+    # scalar all-reduce, one fused scale; shards partition the full
+    # gradient so the result equals the unsharded norm)
+    # ------------------------------------------------------------------
+
+    def clip_grad_norm_(self, max_norm, norm_type=2.0):
+        assert norm_type == 2.0, "only L2 clipping is supported"
+        grads = [
+            u.flat_param.grad for u in self._all_units()
+            if u.flat_param.grad is not None
+        ]
+        if not grads:
+            return torch.zeros((), device=self.device)
+        local = local_sqnorm(grads)
+        if local.device != self.device:
+            local = local.to(self.device)
+        self._comm.all_reduce_scalar_(local)
+        total_norm = local.sqrt()
+        # same formula as torch.nn.utils.clip_grad_norm_
+        clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0)
+        scale_(grads, clip_coef)
+        return total_norm
+
+    # ------------------------------------------------------------------
+    # checkpoint metadata (reference utils.py:29)
+    # ------------------------------------------------------------------
+
+    @staticmethod
+    def _clean_name(name):
+        for tag in ("_fsdp_wrapped_module.", "_checkpoint_wrapped_module."):
+            name = name.replace(tag, "")
+        return name
+
+    def get_shard_metadata(self):
+        """Shard-layout record enabling offline consolidation.
+
+        Returns {"world_size", "rank", "shard_info"} where shard_info
+        maps each flat-param *state_dict key* (raw, as stored in the
+        per-rank checkpoint) to the ordered original-parameter layout
+        of that unit's flat buffer.
+        """
+        shard_info = {}
+        for mod_name, mod in self.named_modules():
+            if not isinstance(mod, FullyShardedDataParallel):
+                continue
+            key = (mod_name + "." if mod_name else "") + "flat_param"
+            inner = getattr(mod, mod.WRAPPER_ATTR)
+            mod_to_prefix = {id(sub): n for n, sub in inner.named_modules()}
+            params = []
+            for m, name, shape, numel, off in mod._param_infos:
+                prefix = mod_to_prefix.get(id(m), "")
+                fqn = (prefix + "." if prefix else "") + name
+                full_fqn = self._clean_name(
+                    (mod_name + "." if mod_name else "") + fqn
+                )
+                params.append(
+                    {
+                        "name": full_fqn,
+                        "shape": list(shape),
+                        "numel": numel,
+                        "offset": off,
+                    }
+                )
+            shard_info[key] = {
+                "params": params,
+                "total_numel": mod._total_numel,
+                "padded_numel": mod._padded_numel,
+            }
+        return {
+            "world_size": self._comm.world_size,
+            "rank": self._comm.rank,
+            "shard_info": shard_info,
+        }
+
+    # state_dict: default nn.Module behavior gives exactly the per-rank
+    # shard checkpoint (flat_param per unit + buffers); keys carry the
+    # wrapper attributes, which consolidation strips via _clean_name.
+
+    def extra_repr(self):
+        return (
+            f"world_size={self._comm.world_size}, rank={self._comm.rank}, "
+            f"total={self._total_numel}, shard={self._shard_numel}, "
+            f"compute_dtype={self.compute_dtype}, "
+            f"reshard_after_forward={self.reshard_after_forward}"
+        )
