@@ -1,0 +1,165 @@
+#!/usr/bin/env python3
+"""Benchmark harness — measures the BASELINE.json headline metric:
+images/sec (whole job) for ViT-10B FSDP training, bs=128 per GPU
+(= global 1024 at 8 GPUs), 224px, synthetic data, bf16 compute with
+fp32 master shards.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`;
+for N>1 the driver launches via torch.distributed.run with one rank per
+GPU over RCCL.  Rank 0 prints ONE JSON line.  Weak scaling: per-GPU
+batch fixed at 128 as N grows.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+MODELS = {
+    # name: (image_size, patch, embed, heads, blocks, mlp_ratio, classes)
+    "vit10b": (224, 14, 5120, 32, 32, 4.0, 1000),
+    "vit-large": (224, 14, 1024, 16, 24, 4.0, 1000),
+    "vit-tiny": (224, 14, 192, 3, 12, 4.0, 1000),
+    "vit60b": (224, 14, 8192, 64, 48, 4.0, 1000),
+}
+PER_GPU_BATCH = 128
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="vit10b", choices=sorted(MODELS))
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--per_gpu_batch", type=int, default=PER_GPU_BATCH)
+    p.add_argument("--shard_on_cpu", action="store_true")
+    p.add_argument("--no_grad_ckpt", action="store_false", dest="grad_ckpt")
+    return p.parse_args()
+
+
+def main():
+    args = parse()
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.cli import parse_args as cli_parse
+    from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
+    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.utils import get_warmup_cosine_scheduler
+
+    device = xdist.init_distributed()
+    world = xdist.get_world_size()
+    rank = xdist.get_rank()
+    n_gpus = world if world > 1 else args.gpus
+    assert n_gpus == world or world == 1, (
+        f"--gpus {args.gpus} but WORLD_SIZE {world}; launch N>1 via torchrun"
+    )
+
+    img, patch, embed, heads, blocks, mlp_ratio, classes = MODELS[args.model]
+    compute_dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    if not torch.cuda.is_available():
+        compute_dtype = torch.float32
+
+    cfg = cli_parse([
+        "--fake_data",
+        "--image_size", str(img), "--patch_size", str(patch),
+        "--embed_dim", str(embed), "--num_heads", str(heads),
+        "--num_blocks", str(blocks), "--mlp_ratio", str(mlp_ratio),
+        "--num_classes", str(classes),
+        "--batch_size", str(args.per_gpu_batch * world),
+    ] + (["--shard_on_cpu"] if args.shard_on_cpu else [])
+      + ([] if args.grad_ckpt else ["--no_grad_ckpt"]))
+
+    torch.manual_seed(1234)
+    t_build = time.time()
+    model = build_fsdp_vit_model(cfg, device, compute_dtype=compute_dtype)
+    loss_fn = CrossEntropyLoss()
+    opt = FusedAdamW(model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay)
+    sched = get_warmup_cosine_scheduler(opt, cfg.warmup_steps, 10**9)
+    xdist.master_print(f"[bench] model built in {time.time() - t_build:.1f}s")
+
+    b = args.per_gpu_batch
+    # synthetic data: random normalized-image-like inputs (NOT zeros — a
+    # zero-filled input inflates clocks/collapses softmax work), random
+    # labels; two rotating buffers staged on device
+    gen = torch.Generator(device="cpu").manual_seed(4242 + rank)
+    batches = []
+    for _ in range(2):
+        x = torch.randn(b, 3, img, img, generator=gen).to(device, compute_dtype)
+        y = torch.randint(0, classes, (b,), generator=gen).to(device)
+        batches.append((x, y))
+
+    def one_step(i):
+        x, y = batches[i % 2]
+        loss = loss_fn(model(x), y)
+        loss.backward()
+        if cfg.clip_grad_norm > 0:
+            model.clip_grad_norm_(cfg.clip_grad_norm)
+        opt.step()
+        sched.step()
+        opt.zero_grad(set_to_none=True)
+        return loss
+
+    def barrier_sync():
+        if xdist.is_distributed():
+            torch.distributed.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        one_step(i)
+    barrier_sync()
+    t0 = time.time()
+    loss = None
+    for i in range(args.steps):
+        loss = one_step(i)
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if xdist.is_distributed():
+        t = t.to(device) if device.type == "cuda" else t
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    images_per_sec = args.steps * b * world / elapsed
+    ms_per_step = elapsed * 1000.0 / args.steps
+    if rank == 0:
+        result = {
+            "metric": (
+                "images/sec (whole node) for ViT-10B bs=1024 224px --fake_data"
+                if args.model == "vit10b"
+                else f"images/sec (whole node), {args.model} FSDP training"
+            ),
+            "value": round(images_per_sec, 2),
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no throughput
+            "dtype": args.dtype if torch.cuda.is_available() else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": {
+                    "vit10b": "ViT-10B (5120d/32h/32L, patch14, 224px)",
+                    "vit-large": "ViT-Large (1024d/16h/24L)",
+                    "vit-tiny": "ViT-Tiny (192d/3h/12L)",
+                    "vit60b": "ViT-60B-class (8192d/64h/48L)",
+                }[args.model],
+                "global_batch": b * world,
+                "seq_len": (img // patch) ** 2,
+                "parallelism": f"fsdp{world}" + ("+cpu_shard" if args.shard_on_cpu else ""),
+                "grad_ckpt": cfg.grad_ckpt,
+                "final_loss": float(loss.item()) if loss is not None else None,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,53 @@
+// Python bindings for the in-tree CDNA4 (gfx950) kernels.
+// Built ahead of time via `python setup.py build_ext --inplace`
+// (PYTORCH_ROCM_ARCH=gfx950); the resulting _C.so lives inside the
+// package so it travels with repo snapshots to GPU boxes.
+
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+
+void fused_adamw(std::vector<torch::Tensor> params,
+                 std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> exp_avgs,
+                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                 double beta1, double beta2, double eps, double weight_decay,
+                 double bias_c1, double bias_c2);
+torch::Tensor multi_tensor_sqnorm(std::vector<torch::Tensor> tensors);
+void multi_tensor_scale(std::vector<torch::Tensor> tensors, double factor);
+void multi_tensor_scale_tensor(std::vector<torch::Tensor> tensors,
+                               torch::Tensor factor);
+
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor target);
+torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
+                                torch::Tensor target, torch::Tensor lse);
+
+std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, double scale);
+std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    double scale);
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16, fp32 stats)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward");
+  m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (fp32)");
+  m.def("multi_tensor_sqnorm", &multi_tensor_sqnorm,
+        "sum of squares over tensor list");
+  m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scalar scale");
+  m.def("multi_tensor_scale_tensor", &multi_tensor_scale_tensor,
+        "in-place scale by device scalar");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused softmax CE forward");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused softmax CE backward");
+  m.def("fmha_fwd", &fmha_fwd, "flash attention forward (bf16, head_dim<=192)");
+  m.def("fmha_bwd", &fmha_bwd, "flash attention backward (recompute-based)");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+}

@@ -1,0 +1,363 @@
+// Flash-style fused multi-head attention for CDNA4 (kernel K4,
+// SURVEY.md §2D): per-head softmax(Q K^T * scale) V with online softmax,
+// O(T) memory, for the ViT-10B head_dim of 160 (= 5120/32, reference
+// run_vit_training.py:341-342) — off the usual 64/128 fast paths, so the
+// tiling is D-generic over 32-wide MFMA K-chunks (160 = 5 x 32).
+//
+// Structure (forward):
+//   grid = (ceil(T/64), B*H); block = 256 threads = 4 waves.
+//   Wave w owns 16 query rows.  Per 32-key tile:
+//     - K tile [32][Dpad] and transposed V tile [D][32] staged in LDS
+//       (padded row strides, conflict-checked for ds_read_b128 groups),
+//     - S = Q K^T via __builtin_amdgcn_mfma_f32_16x16x32_bf16
+//       (2 k-subtiles x D/32 chunks), fp32 accumulators,
+//     - online softmax in registers (row state lives in the C-fragment
+//       lane groups; cross-lane row reduce = 4 x shfl_xor over the
+//       16-lane column group),
+//     - P routed through a small per-wave LDS tile to re-shape the
+//       softmax output (C layout) into the MFMA A layout for P V.
+//   Epilogue: O /= rowsum, store bf16, write LSE (fp32) for backward.
+//
+// Backward: recompute-based composition — the batched GEMMs (S, dP, dV,
+// dQ, dK) go through hipBLASLt (at::matmul), while the softmax-gradient
+// elementwise work (P = exp(S*scale - lse), dS = scale*P*(dP - Delta))
+// and the Delta = rowsum(dO*O) reduction are fused custom kernels.
+//
+// MFMA fragment layout assumptions (verified on-device by mfma_probe,
+// tests/test_gpu_kernels.py):
+//   A[i][k]: lane l holds a[j] = A[l&15][(l>>4)*8 + j]
+//   B[k][j]: lane l holds b[j] = B[(l>>4)*8 + j][l&15]
+//   C[i][j]: lane l holds c[r] = C[(l>>4)*4 + r][l&15]
+
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int kBlockThreads = 256;
+constexpr int kQTile = 64;   // q rows per workgroup (16 per wave)
+constexpr int kKTile = 32;   // k rows per inner iteration
+constexpr int kMaxD = 192;   // supports head_dim up to 192 (10B uses 160)
+constexpr int kMaxDP = 192;  // padded to 32
+// LDS row strides in elements, padded so the 16-lane ds_read_b128 column
+// groups land on distinct banks (stride*2B must stay 16B-aligned)
+constexpr int kKStride = kMaxDP + 8;
+constexpr int kVStride = kKTile + 8;  // 40
+constexpr int kPStride = kKTile + 8;  // 40
+
+__device__ __forceinline__ float lane_bf(short s) {
+  return bf16_to_f32((unsigned short)s);
+}
+
+struct SharedMem {
+  short k_tile[kKTile][kKStride];
+  short v_tile[kMaxD][kVStride];  // transposed: [d][k]
+  short p_tile[4][16][kPStride];  // per-wave P re-layout buffer
+};
+
+__global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, short* __restrict__ o,
+    float* __restrict__ lse_out, int T, int D, int DP, float scale) {
+  HIP_DYNAMIC_SHARED(char, smem_raw)
+  SharedMem& sm = *reinterpret_cast<SharedMem*>(smem_raw);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int col = lane & 15;        // fragment column group
+  const int seg = lane >> 4;        // fragment k/row segment (0..3)
+  const long bh = blockIdx.y;
+  const int q_base = blockIdx.x * kQTile;
+  const int q_row0 = q_base + wave * 16;  // this wave's first q row
+
+  const long qkv_off = bh * (long)T * D;
+  const int nkc = DP / 32;  // QK^T K-chunks
+  const int nc = D / 16;    // PV / O column chunks
+
+  // ---- load Q fragments to registers: chunk kc, elem j ->
+  // Q[q_row0 + col][seg*8 + j + 32*kc] (zero-padded beyond D / T) ----
+  bf16x8 q_frag[kMaxDP / 32];
+  {
+    const int q_row = q_row0 + col;
+    const bool valid = q_row < T;
+    const long base = qkv_off + (long)q_row * D;
+    for (int kc = 0; kc < nkc; ++kc) {
+      const int d0 = kc * 32 + seg * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        q_frag[kc][j] =
+            (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
+      }
+    }
+  }
+
+  // ---- online softmax state (per lane: 4 q rows of this wave) ----
+  float m_run[4], l_run[4];
+  f32x4 o_acc[kMaxD / 16];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -INFINITY;
+    l_run[r] = 0.f;
+  }
+  for (int c = 0; c < nc; ++c) o_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_ktiles = (T + kKTile - 1) / kKTile;
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    const int k_base = kt * kKTile;
+    __syncthreads();  // previous iteration's LDS reads are done
+
+    // ---- cooperative K tile load: [32][D] -> sm.k_tile (16B vectors) ----
+    {
+      const int vecs_per_row = DP / 8;
+      const int total = kKTile * vecs_per_row;
+      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
+        const int kr = i / vecs_per_row;
+        const int dc = (i % vecs_per_row) * 8;
+        bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+        const int k_row = k_base + kr;
+        if (k_row < T && dc < D) {
+          val = *reinterpret_cast<const bf16x8*>(&k[qkv_off + (long)k_row * D + dc]);
+        }
+        *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = val;
+      }
+    }
+    // ---- V tile load, transposed into [d][k] ----
+    {
+      const int vecs_per_row = D / 8;
+      const int total = kKTile * vecs_per_row;
+      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
+        const int kr = i / vecs_per_row;
+        const int dc = (i % vecs_per_row) * 8;
+        const int k_row = k_base + kr;
+        bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (k_row < T) {
+          val = *reinterpret_cast<const bf16x8*>(&v[qkv_off + (long)k_row * D + dc]);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) sm.v_tile[dc + j][kr] = val[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- S = scale * Q K^T for this wave: [16 q][32 k], 2 k-subtiles ----
+    f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll 2
+    for (int kk = 0; kk < 2; ++kk) {
+      for (int kc = 0; kc < nkc; ++kc) {
+        // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
+        bf16x8 b_frag =
+            *reinterpret_cast<const bf16x8*>(&sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
+        s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            q_frag[kc], b_frag, s_frag[kk], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax update ----
+    float p_val[2][4];
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float s0 = s_frag[0][r] * scale;
+      float s1 = s_frag[1][r] * scale;
+      if (k_base + col >= T) s0 = -INFINITY;
+      if (k_base + 16 + col >= T) s1 = -INFINITY;
+      float m_tile = fmaxf(s0, s1);
+      // row max across the 16-lane column group
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        m_tile = fmaxf(m_tile, __shfl_xor(m_tile, off));
+      const float m_new = fmaxf(m_run[r], m_tile);
+      alpha[r] = __expf(m_run[r] - m_new);  // exp(-inf - finite) = 0
+      m_run[r] = m_new;
+      float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
+      float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
+      p_val[0][r] = p0;
+      p_val[1][r] = p1;
+      float row_sum = p0 + p1;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        row_sum += __shfl_xor(row_sum, off);
+      l_run[r] = l_run[r] * alpha[r] + row_sum;
+    }
+
+    // rescale O
+#pragma unroll
+    for (int c = 0; c < kMaxD / 16; ++c) {
+      if (c >= nc) break;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[c][r] *= alpha[r];
+    }
+
+    // ---- re-layout P (C layout) -> A layout via per-wave LDS tile ----
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        sm.p_tile[wave][seg * 4 + r][kk * 16 + col] =
+            (short)f32_to_bf16(p_val[kk][r]);
+      }
+    }
+    // same-wave LDS visibility: wait LDS ops, pin the scheduler
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    bf16x8 p_frag =
+        *reinterpret_cast<const bf16x8*>(&sm.p_tile[wave][col][seg * 8]);
+
+    // ---- O += P V : nc chunks of 16 output columns ----
+    for (int c = 0; c < nc; ++c) {
+      // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
+      bf16x8 v_frag =
+          *reinterpret_cast<const bf16x8*>(&sm.v_tile[c * 16 + col][seg * 8]);
+      o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
+                                                         o_acc[c], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: normalize, store O and LSE ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int q_row = q_row0 + seg * 4 + r;
+    if (q_row >= T) continue;
+    const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    const long out_base = qkv_off + (long)q_row * D;
+    for (int c = 0; c < nc; ++c) {
+      o[out_base + c * 16 + col] = (short)f32_to_bf16(o_acc[c][r] * inv_l);
+    }
+    if (col == 0) {
+      lse_out[bh * T + q_row] = m_run[r] + __logf(l_run[r]);
+    }
+  }
+}
+
+// Delta = rowsum(dO * O): one wave per row
+__global__ __launch_bounds__(64) void fmha_rowdot_kernel(
+    const unsigned short* __restrict__ dout, const unsigned short* __restrict__ o,
+    float* __restrict__ delta, int D) {
+  const long row = blockIdx.x;
+  const unsigned short* dr = dout + row * D;
+  const unsigned short* orow = o + row * D;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < D; i += 64)
+    acc += bf16_to_f32(dr[i]) * bf16_to_f32(orow[i]);
+  acc = wave_all_sum(acc);
+  if (threadIdx.x == 0) delta[row] = acc;
+}
+
+// P = exp(scale*S - lse); dS = scale * P * (dP - Delta)   (fused, bf16 io)
+__global__ __launch_bounds__(256) void fmha_dsoftmax_kernel(
+    const unsigned short* __restrict__ s_raw,
+    const unsigned short* __restrict__ dp, const float* __restrict__ lse,
+    const float* __restrict__ delta, unsigned short* __restrict__ p_out,
+    unsigned short* __restrict__ ds_out, int T, float scale) {
+  const long row = blockIdx.x;  // over B*H*T rows
+  const float row_lse = lse[row];
+  const float row_delta = delta[row];
+  const long base = row * T;
+  for (int i = threadIdx.x; i < T; i += 256) {
+    const float sv = bf16_to_f32(s_raw[base + i]) * scale;
+    const float p = __expf(sv - row_lse);
+    const float d = scale * p * (bf16_to_f32(dp[base + i]) - row_delta);
+    p_out[base + i] = f32_to_bf16(p);
+    ds_out[base + i] = f32_to_bf16(d);
+  }
+}
+
+// Layout probe: one wave computes a single 16x16x32 MFMA from row-major
+// fp32 A[16][32], B[32][16] using the documented fragment maps; the GPU
+// test compares against torch.matmul to pin the layout assumptions.
+__global__ __launch_bounds__(64) void mfma_probe_kernel(
+    const float* __restrict__ a, const float* __restrict__ b,
+    float* __restrict__ c) {
+  const int lane = threadIdx.x & 63;
+  const int col = lane & 15, seg = lane >> 4;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = (short)f32_to_bf16(a[col * 32 + seg * 8 + j]);
+    bf[j] = (short)f32_to_bf16(b[(seg * 8 + j) * 16 + col]);
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) c[(seg * 4 + r) * 16 + col] = acc[r];
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "fmha_fwd: bf16 only");
+  TORCH_CHECK(q.dim() == 4, "fmha_fwd: [B, H, T, D]");
+  const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
+  TORCH_CHECK(D % 16 == 0 && D <= kMaxD,
+              "fmha_fwd: head_dim must be a multiple of 16 and <= ", kMaxD);
+  const int DP = ((D + 31) / 32) * 32;
+
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
+  dim3 grid((T + kQTile - 1) / kQTile, B * H);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(fmha_fwd_kernel, grid, dim3(kBlockThreads),
+                     sizeof(SharedMem), stream, (const short*)q.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     (short*)o.data_ptr(), lse.data_ptr<float>(), T, D, DP,
+                     (float)scale);
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    double scale) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+  const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  auto delta = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(fmha_rowdot_kernel, dim3(B * H * T), dim3(64), 0, stream,
+                     (const unsigned short*)dout.data_ptr(),
+                     (const unsigned short*)o.data_ptr(),
+                     delta.data_ptr<float>(), D);
+  HIP_CHECK_LAST();
+
+  // recompute raw scores and dP with hipBLASLt batched GEMMs
+  auto s_raw = at::matmul(q, k.transpose(-2, -1));
+  auto dp = at::matmul(dout, v.transpose(-2, -1));
+
+  auto p = torch::empty_like(s_raw);
+  auto ds = torch::empty_like(s_raw);
+  hipLaunchKernelGGL(fmha_dsoftmax_kernel, dim3(B * H * T), dim3(256), 0,
+                     stream, (const unsigned short*)s_raw.data_ptr(),
+                     (const unsigned short*)dp.data_ptr(),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     (unsigned short*)p.data_ptr(),
+                     (unsigned short*)ds.data_ptr(), T, (float)scale);
+  HIP_CHECK_LAST();
+
+  auto dv = at::matmul(p.transpose(-2, -1), dout);
+  auto dq = at::matmul(ds, k);
+  auto dk = at::matmul(ds.transpose(-2, -1), q);
+  return {dq, dk, dv};
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(b.is_cuda() && b.sizes() == torch::IntArrayRef({32, 16}));
+  auto a32 = a.to(torch::kFloat32).contiguous();
+  auto b32 = b.to(torch::kFloat32).contiguous();
+  auto c = torch::empty({16, 16}, a32.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     a32.data_ptr<float>(), b32.data_ptr<float>(),
+                     c.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return c;
+}

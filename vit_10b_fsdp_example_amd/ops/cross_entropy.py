@@ -30,7 +30,7 @@ class _CrossEntropyFn(torch.autograd.Function):
 
 def cross_entropy(logits, target):
     """Mean-reduced cross entropy on [N, C] logits, [N] int64 targets."""
-    if use_hip(logits):
+    if logits.dtype == torch.bfloat16 and use_hip(logits):
         return _CrossEntropyFn.apply(logits.contiguous(), target.contiguous())
     return F.cross_entropy(logits.float(), target)
 

@@ -165,9 +165,11 @@ class FullyShardedDataParallel(nn.Module):
         )
         _free_storage(self._full_grad)
 
-        # runtime state
+        # runtime state (note: _root_ref may hold an nn.Module and must
+        # NOT go through nn.Module.__setattr__, which would register it
+        # as a child and create a module cycle)
         self._is_root = None
-        self._root_ref = None
+        object.__setattr__(self, "_root_ref", None)
         self._fwd_order = None  # root only: units in forward execution order
         self._pre_bwd_done = False
         self._grads_arrived = 0
@@ -228,7 +230,7 @@ class FullyShardedDataParallel(nn.Module):
             if self._is_root:
                 self._fwd_order = []
         root = _EXEC_STACK[0] if _EXEC_STACK else self
-        self._root_ref = root
+        object.__setattr__(self, "_root_ref", root)
         if root._fwd_order is not None and self not in root._fwd_order:
             root._fwd_order.append(self)
 
