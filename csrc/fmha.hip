@@ -4,6 +4,12 @@
 // run_vit_training.py:341-342) — off the usual 64/128 fast paths, so the
 // tiling is D-generic over 32-wide MFMA K-chunks (160 = 5 x 32).
 //
+// head_dim is a TEMPLATE parameter: every fragment/accumulator loop is
+// compile-time bounded so the f32x4 MFMA accumulators stay in VGPRs — a
+// runtime-indexed ext_vector array silently spills to scratch, which
+// measured ~300x slower (guide §5.4 rule 20; seen in the first profile,
+// profiles/).
+//
 // Structure (forward):
 //   grid = (ceil(T/64), B*H); block = 256 threads = 4 waves.
 //   Wave w owns 16 query rows.  Per 32-key tile:
@@ -23,8 +29,8 @@
 // elementwise work (P = exp(S*scale - lse), dS = scale*P*(dP - Delta))
 // and the Delta = rowsum(dO*O) reduction are fused custom kernels.
 //
-// MFMA fragment layout assumptions (verified on-device by mfma_probe,
-// tests/test_gpu_kernels.py):
+// MFMA fragment layout (verified on-device by mfma_probe,
+// tests/test_gpu_kernels.py::test_mfma_probe_layout):
 //   A[i][k]: lane l holds a[j] = A[l&15][(l>>4)*8 + j]
 //   B[k][j]: lane l holds b[j] = B[(l>>4)*8 + j][l&15]
 //   C[i][j]: lane l holds c[r] = C[(l>>4)*4 + r][l&15]
@@ -40,71 +46,74 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr int kBlockThreads = 256;
-constexpr int kQTile = 64;   // q rows per workgroup (16 per wave)
-constexpr int kKTile = 32;   // k rows per inner iteration
-constexpr int kMaxD = 192;   // supports head_dim up to 192 (10B uses 160)
-constexpr int kMaxDP = 192;  // padded to 32
-// LDS row strides in elements, padded so the 16-lane ds_read_b128 column
-// groups land on distinct banks (stride*2B must stay 16B-aligned)
-constexpr int kKStride = kMaxDP + 8;
-constexpr int kVStride = kKTile + 8;  // 40
-constexpr int kPStride = kKTile + 8;  // 40
+constexpr int kQTile = 64;  // q rows per workgroup (16 per wave)
+constexpr int kKTile = 32;  // k rows per inner iteration
+constexpr int kMaxD = 192;  // supports head_dim up to 192 (10B uses 160)
 
-__device__ __forceinline__ float lane_bf(short s) {
-  return bf16_to_f32((unsigned short)s);
-}
-
-struct SharedMem {
-  short k_tile[kKTile][kKStride];
-  short v_tile[kMaxD][kVStride];  // transposed: [d][k]
-  short p_tile[4][16][kPStride];  // per-wave P re-layout buffer
+template <int D>
+struct FmhaShapes {
+  static constexpr int DP = ((D + 31) / 32) * 32;  // zero-padded for QK
+  static constexpr int NKC = DP / 32;              // QK^T K-chunks
+  static constexpr int NC = D / 16;                // PV / O column chunks
+  // LDS row strides in elements, padded so the 16-lane ds_read_b128
+  // column groups land on distinct banks (stride*2B stays 16B-aligned)
+  static constexpr int KStride = DP + 8;
+  static constexpr int VStride = kKTile + 8;  // 40
+  static constexpr int PStride = kKTile + 8;  // 40
+  struct Shared {
+    short k_tile[kKTile][KStride];
+    short v_tile[D][VStride];       // transposed: [d][k]
+    short p_tile[4][16][PStride];   // per-wave P re-layout buffer
+  };
 };
 
+template <int D>
 __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse_out, int T, int D, int DP, float scale) {
+    float* __restrict__ lse_out, int T, float scale) {
+  using S = FmhaShapes<D>;
+  constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   HIP_DYNAMIC_SHARED(char, smem_raw)
-  SharedMem& sm = *reinterpret_cast<SharedMem*>(smem_raw);
+  typename S::Shared& sm = *reinterpret_cast<typename S::Shared*>(smem_raw);
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int col = lane & 15;        // fragment column group
-  const int seg = lane >> 4;        // fragment k/row segment (0..3)
+  const int col = lane & 15;  // fragment column group
+  const int seg = lane >> 4;  // fragment k/row segment (0..3)
   const long bh = blockIdx.y;
   const int q_base = blockIdx.x * kQTile;
   const int q_row0 = q_base + wave * 16;  // this wave's first q row
 
   const long qkv_off = bh * (long)T * D;
-  const int nkc = DP / 32;  // QK^T K-chunks
-  const int nc = D / 16;    // PV / O column chunks
 
   // ---- load Q fragments to registers: chunk kc, elem j ->
   // Q[q_row0 + col][seg*8 + j + 32*kc] (zero-padded beyond D / T) ----
-  bf16x8 q_frag[kMaxDP / 32];
+  bf16x8 q_frag[NKC];
   {
     const int q_row = q_row0 + col;
     const bool valid = q_row < T;
     const long base = qkv_off + (long)q_row * D;
-    for (int kc = 0; kc < nkc; ++kc) {
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        q_frag[kc][j] =
-            (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
+        q_frag[kc][j] = (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
       }
     }
   }
 
   // ---- online softmax state (per lane: 4 q rows of this wave) ----
   float m_run[4], l_run[4];
-  f32x4 o_acc[kMaxD / 16];
+  f32x4 o_acc[NC];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     m_run[r] = -INFINITY;
     l_run[r] = 0.f;
   }
-  for (int c = 0; c < nc; ++c) o_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int c = 0; c < NC; ++c) o_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int n_ktiles = (T + kKTile - 1) / kKTile;
   for (int kt = 0; kt < n_ktiles; ++kt) {
@@ -113,30 +122,32 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
 
     // ---- cooperative K tile load: [32][D] -> sm.k_tile (16B vectors) ----
     {
-      const int vecs_per_row = DP / 8;
-      const int total = kKTile * vecs_per_row;
+      constexpr int vecs_per_row = DP / 8;
+      constexpr int total = kKTile * vecs_per_row;
       for (int i = threadIdx.x; i < total; i += kBlockThreads) {
         const int kr = i / vecs_per_row;
         const int dc = (i % vecs_per_row) * 8;
         bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
         const int k_row = k_base + kr;
         if (k_row < T && dc < D) {
-          val = *reinterpret_cast<const bf16x8*>(&k[qkv_off + (long)k_row * D + dc]);
+          val = *reinterpret_cast<const bf16x8*>(
+              &k[qkv_off + (long)k_row * D + dc]);
         }
         *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = val;
       }
     }
     // ---- V tile load, transposed into [d][k] ----
     {
-      const int vecs_per_row = D / 8;
-      const int total = kKTile * vecs_per_row;
+      constexpr int vecs_per_row = D / 8;
+      constexpr int total = kKTile * vecs_per_row;
       for (int i = threadIdx.x; i < total; i += kBlockThreads) {
         const int kr = i / vecs_per_row;
         const int dc = (i % vecs_per_row) * 8;
         const int k_row = k_base + kr;
         bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
         if (k_row < T) {
-          val = *reinterpret_cast<const bf16x8*>(&v[qkv_off + (long)k_row * D + dc]);
+          val = *reinterpret_cast<const bf16x8*>(
+              &v[qkv_off + (long)k_row * D + dc]);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) sm.v_tile[dc + j][kr] = val[j];
@@ -146,12 +157,13 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
 
     // ---- S = scale * Q K^T for this wave: [16 q][32 k], 2 k-subtiles ----
     f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-#pragma unroll 2
+#pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      for (int kc = 0; kc < nkc; ++kc) {
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
         // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
-        bf16x8 b_frag =
-            *reinterpret_cast<const bf16x8*>(&sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
         s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             q_frag[kc], b_frag, s_frag[kk], 0, 0, 0);
       }
@@ -187,8 +199,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
 
     // rescale O
 #pragma unroll
-    for (int c = 0; c < kMaxD / 16; ++c) {
-      if (c >= nc) break;
+    for (int c = 0; c < NC; ++c) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[c][r] *= alpha[r];
     }
@@ -208,8 +219,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     bf16x8 p_frag =
         *reinterpret_cast<const bf16x8*>(&sm.p_tile[wave][col][seg * 8]);
 
-    // ---- O += P V : nc chunks of 16 output columns ----
-    for (int c = 0; c < nc; ++c) {
+    // ---- O += P V : NC chunks of 16 output columns ----
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
       // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
       bf16x8 v_frag =
           *reinterpret_cast<const bf16x8*>(&sm.v_tile[c * 16 + col][seg * 8]);
@@ -225,7 +237,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     if (q_row >= T) continue;
     const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
     const long out_base = qkv_off + (long)q_row * D;
-    for (int c = 0; c < nc; ++c) {
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
       o[out_base + c * 16 + col] = (short)f32_to_bf16(o_acc[c][r] * inv_l);
     }
     if (col == 0) {
@@ -236,8 +249,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
 
 // Delta = rowsum(dO * O): one wave per row
 __global__ __launch_bounds__(64) void fmha_rowdot_kernel(
-    const unsigned short* __restrict__ dout, const unsigned short* __restrict__ o,
-    float* __restrict__ delta, int D) {
+    const unsigned short* __restrict__ dout,
+    const unsigned short* __restrict__ o, float* __restrict__ delta, int D) {
   const long row = blockIdx.x;
   const unsigned short* dr = dout + row * D;
   const unsigned short* orow = o + row * D;
@@ -248,7 +261,9 @@ __global__ __launch_bounds__(64) void fmha_rowdot_kernel(
   if (threadIdx.x == 0) delta[row] = acc;
 }
 
-// P = exp(scale*S - lse); dS = scale * P * (dP - Delta)   (fused, bf16 io)
+// P = exp(scale*S - lse); dS = scale * P * (dP - Delta)
+// fused, bf16 io, 8-wide vectorized (rows are T elements, T % 8 == 0 on
+// every ViT config; scalar fallback otherwise)
 __global__ __launch_bounds__(256) void fmha_dsoftmax_kernel(
     const unsigned short* __restrict__ s_raw,
     const unsigned short* __restrict__ dp, const float* __restrict__ lse,
@@ -258,12 +273,30 @@ __global__ __launch_bounds__(256) void fmha_dsoftmax_kernel(
   const float row_lse = lse[row];
   const float row_delta = delta[row];
   const long base = row * T;
-  for (int i = threadIdx.x; i < T; i += 256) {
-    const float sv = bf16_to_f32(s_raw[base + i]) * scale;
-    const float p = __expf(sv - row_lse);
-    const float d = scale * p * (bf16_to_f32(dp[base + i]) - row_delta);
-    p_out[base + i] = f32_to_bf16(p);
-    ds_out[base + i] = f32_to_bf16(d);
+  if ((T & 7) == 0) {
+    const int nvec = T / 8;
+    const ushort8_t* sv = (const ushort8_t*)(s_raw + base);
+    const ushort8_t* dpv = (const ushort8_t*)(dp + base);
+    ushort8_t* pv = (ushort8_t*)(p_out + base);
+    ushort8_t* dsv = (ushort8_t*)(ds_out + base);
+    for (int i = threadIdx.x; i < nvec; i += 256) {
+      ushort8_t s8 = sv[i], d8 = dpv[i], p8, ds8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float p = __expf(bf16_to_f32(s8[j]) * scale - row_lse);
+        p8[j] = f32_to_bf16(p);
+        ds8[j] = f32_to_bf16(scale * p * (bf16_to_f32(d8[j]) - row_delta));
+      }
+      pv[i] = p8;
+      dsv[i] = ds8;
+    }
+  } else {
+    for (int i = threadIdx.x; i < T; i += 256) {
+      const float p = __expf(bf16_to_f32(s_raw[base + i]) * scale - row_lse);
+      p_out[base + i] = f32_to_bf16(p);
+      ds_out[base + i] =
+          f32_to_bf16(scale * p * (bf16_to_f32(dp[base + i]) - row_delta));
+    }
   }
 }
 
@@ -287,6 +320,20 @@ __global__ __launch_bounds__(64) void mfma_probe_kernel(
   for (int r = 0; r < 4; ++r) c[(seg * 4 + r) * 16 + col] = acc[r];
 }
 
+template <int D>
+void launch_fmha_fwd(const torch::Tensor& q, const torch::Tensor& k,
+                     const torch::Tensor& v, torch::Tensor& o,
+                     torch::Tensor& lse, int T, long BH, float scale) {
+  dim3 grid((T + kQTile - 1) / kQTile, (unsigned)BH);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(fmha_fwd_kernel<D>, grid, dim3(kBlockThreads),
+                     sizeof(typename FmhaShapes<D>::Shared), stream,
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                     (const short*)v.data_ptr(), (short*)o.data_ptr(),
+                     lse.data_ptr<float>(), T, scale);
+  HIP_CHECK_LAST();
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
@@ -298,18 +345,27 @@ std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
   const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
   TORCH_CHECK(D % 16 == 0 && D <= kMaxD,
               "fmha_fwd: head_dim must be a multiple of 16 and <= ", kMaxD);
-  const int DP = ((D + 31) / 32) * 32;
 
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
-  dim3 grid((T + kQTile - 1) / kQTile, B * H);
-  auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(fmha_fwd_kernel, grid, dim3(kBlockThreads),
-                     sizeof(SharedMem), stream, (const short*)q.data_ptr(),
-                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     (short*)o.data_ptr(), lse.data_ptr<float>(), T, D, DP,
-                     (float)scale);
-  HIP_CHECK_LAST();
+  const long BH = (long)B * H;
+  const float s = (float)scale;
+  switch (D) {
+    case 16: launch_fmha_fwd<16>(q, k, v, o, lse, T, BH, s); break;
+    case 32: launch_fmha_fwd<32>(q, k, v, o, lse, T, BH, s); break;
+    case 48: launch_fmha_fwd<48>(q, k, v, o, lse, T, BH, s); break;
+    case 64: launch_fmha_fwd<64>(q, k, v, o, lse, T, BH, s); break;
+    case 80: launch_fmha_fwd<80>(q, k, v, o, lse, T, BH, s); break;
+    case 96: launch_fmha_fwd<96>(q, k, v, o, lse, T, BH, s); break;
+    case 112: launch_fmha_fwd<112>(q, k, v, o, lse, T, BH, s); break;
+    case 128: launch_fmha_fwd<128>(q, k, v, o, lse, T, BH, s); break;
+    case 144: launch_fmha_fwd<144>(q, k, v, o, lse, T, BH, s); break;
+    case 160: launch_fmha_fwd<160>(q, k, v, o, lse, T, BH, s); break;
+    case 176: launch_fmha_fwd<176>(q, k, v, o, lse, T, BH, s); break;
+    case 192: launch_fmha_fwd<192>(q, k, v, o, lse, T, BH, s); break;
+    default:
+      TORCH_CHECK(false, "fmha_fwd: unsupported head_dim ", D);
+  }
   return {o, lse};
 }
 

@@ -185,14 +185,13 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                      1.f / d);
   HIP_CHECK_LAST();
 
-  // pick row chunking so the partial grid fills the 256 CUs
-  int rows_per_chunk = 512;
-  int n_chunks = (int)((n + rows_per_chunk - 1) / rows_per_chunk);
-  if (n_chunks < 8 && n >= 8) {
-    n_chunks = 8;
-    rows_per_chunk = (int)((n + n_chunks - 1) / n_chunks);
-    n_chunks = (int)((n + rows_per_chunk - 1) / rows_per_chunk);
-  }
+  // pick row chunking so the partial grid fills the chip: target ~2048
+  // blocks total (256 CUs x 8, guide G11)
+  const int grid_x = (d + kBlock - 1) / kBlock;
+  int n_chunks = std::max(1, 2048 / std::max(grid_x, 1));
+  int rows_per_chunk = (int)((n + n_chunks - 1) / n_chunks);
+  rows_per_chunk = std::max(rows_per_chunk, 16);
+  n_chunks = (int)((n + rows_per_chunk - 1) / rows_per_chunk);
   auto opts = x.options().dtype(torch::kFloat32);
   auto dw_part = torch::empty({n_chunks, d}, opts);
   auto db_part = torch::empty({n_chunks, d}, opts);
