@@ -48,6 +48,9 @@ def main():
     from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
     from vit_10b_fsdp_example_amd.utils import get_warmup_cosine_scheduler
 
+    from vit_10b_fsdp_example_amd.tuning import enable_tunableop
+
+    enable_tunableop()
     device = xdist.init_distributed()
     world = xdist.get_world_size()
     rank = xdist.get_rank()

@@ -47,7 +47,8 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr int kBlockThreads = 256;
-constexpr int kQTile = 64;  // q rows per workgroup (16 per wave)
+constexpr int kQSub = 2;    // 16-row q sub-tiles per wave
+constexpr int kQTile = 64 * kQSub;  // q rows per workgroup (32 per wave)
 constexpr int kKTile = 32;  // k rows per inner iteration
 constexpr int kMaxD = 192;  // supports head_dim up to 192 (10B uses 160)
 
@@ -63,11 +64,14 @@ struct FmhaShapes {
   static constexpr int PStride = kKTile + 8;  // 40
   struct Shared {
     short k_tile[kKTile][KStride];
-    short v_tile[D][VStride];       // transposed: [d][k]
-    short p_tile[4][16][PStride];   // per-wave P re-layout buffer
+    short v_tile[D][VStride];             // transposed: [d][k]
+    short p_tile[4][kQSub][16][PStride];  // per-wave P re-layout buffers
   };
 };
 
+// grid = (B*H, ceil(T/kQTile)): bh on x so a head's q-tiles share the
+// XCD whose L2 already holds its K/V (blockIdx linearization round-
+// robins x across XCDs; y strides keep bh%8 constant -> same XCD)
 template <int D>
 __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
@@ -82,17 +86,19 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
   const int wave = threadIdx.x >> 6;
   const int col = lane & 15;  // fragment column group
   const int seg = lane >> 4;  // fragment k/row segment (0..3)
-  const long bh = blockIdx.y;
-  const int q_base = blockIdx.x * kQTile;
-  const int q_row0 = q_base + wave * 16;  // this wave's first q row
+  const long bh = blockIdx.x;
+  const int q_base = blockIdx.y * kQTile;
+  // this wave's first q row, per sub-tile qs: q_row0 + 16*qs
+  const int q_row0 = q_base + wave * (16 * kQSub);
 
   const long qkv_off = bh * (long)T * D;
 
-  // ---- load Q fragments to registers: chunk kc, elem j ->
-  // Q[q_row0 + col][seg*8 + j + 32*kc] (zero-padded beyond D / T) ----
-  bf16x8 q_frag[NKC];
-  {
-    const int q_row = q_row0 + col;
+  // ---- load Q fragments to registers: sub-tile qs, chunk kc, elem j ->
+  // Q[q_row0 + 16*qs + col][seg*8 + j + 32*kc] (zero-padded beyond D/T) --
+  bf16x8 q_frag[kQSub][NKC];
+#pragma unroll
+  for (int qs = 0; qs < kQSub; ++qs) {
+    const int q_row = q_row0 + 16 * qs + col;
     const bool valid = q_row < T;
     const long base = qkv_off + (long)q_row * D;
 #pragma unroll
@@ -100,21 +106,25 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
       const int d0 = kc * 32 + seg * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        q_frag[kc][j] = (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
+        q_frag[qs][kc][j] =
+            (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
       }
     }
   }
 
-  // ---- online softmax state (per lane: 4 q rows of this wave) ----
-  float m_run[4], l_run[4];
-  f32x4 o_acc[NC];
+  // ---- online softmax state (per lane: 4 q rows per sub-tile) ----
+  float m_run[kQSub][4], l_run[kQSub][4];
+  f32x4 o_acc[kQSub][NC];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_run[r] = -INFINITY;
-    l_run[r] = 0.f;
+  for (int qs = 0; qs < kQSub; ++qs) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_run[qs][r] = -INFINITY;
+      l_run[qs][r] = 0.f;
+    }
+#pragma unroll
+    for (int c = 0; c < NC; ++c) o_acc[qs][c] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
-#pragma unroll
-  for (int c = 0; c < NC; ++c) o_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int n_ktiles = (T + kKTile - 1) / kKTile;
   for (int kt = 0; kt < n_ktiles; ++kt) {
@@ -156,94 +166,104 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- S = scale * Q K^T for this wave: [16 q][32 k], 2 k-subtiles ----
-    f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+    // ---- per q sub-tile: S = scale*QK^T, online softmax, stage P ----
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int qs = 0; qs < kQSub; ++qs) {
+      f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
-      for (int kc = 0; kc < NKC; ++kc) {
-        // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
-        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-            &sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
-        s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            q_frag[kc], b_frag, s_frag[kk], 0, 0, 0);
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int kc = 0; kc < NKC; ++kc) {
+          // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
+          bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+              &sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
+          s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              q_frag[qs][kc], b_frag, s_frag[kk], 0, 0, 0);
+        }
       }
-    }
 
-    // ---- online softmax update ----
-    float p_val[2][4];
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float s0 = s_frag[0][r] * scale;
-      float s1 = s_frag[1][r] * scale;
-      if (k_base + col >= T) s0 = -INFINITY;
-      if (k_base + 16 + col >= T) s1 = -INFINITY;
-      float m_tile = fmaxf(s0, s1);
-      // row max across the 16-lane column group
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        m_tile = fmaxf(m_tile, __shfl_xor(m_tile, off));
-      const float m_new = fmaxf(m_run[r], m_tile);
-      alpha[r] = __expf(m_run[r] - m_new);  // exp(-inf - finite) = 0
-      m_run[r] = m_new;
-      float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
-      float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
-      p_val[0][r] = p0;
-      p_val[1][r] = p1;
-      float row_sum = p0 + p1;
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        row_sum += __shfl_xor(row_sum, off);
-      l_run[r] = l_run[r] * alpha[r] + row_sum;
-    }
-
-    // rescale O
-#pragma unroll
-    for (int c = 0; c < NC; ++c) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[c][r] *= alpha[r];
-    }
-
-    // ---- re-layout P (C layout) -> A layout via per-wave LDS tile ----
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+      float p_val[2][4];
+      float alpha[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        sm.p_tile[wave][seg * 4 + r][kk * 16 + col] =
-            (short)f32_to_bf16(p_val[kk][r]);
+        float s0 = s_frag[0][r] * scale;
+        float s1 = s_frag[1][r] * scale;
+        if (k_base + col >= T) s0 = -INFINITY;
+        if (k_base + 16 + col >= T) s1 = -INFINITY;
+        float m_tile = fmaxf(s0, s1);
+        // row max across the 16-lane column group
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          m_tile = fmaxf(m_tile, __shfl_xor(m_tile, off));
+        const float m_new = fmaxf(m_run[qs][r], m_tile);
+        alpha[r] = __expf(m_run[qs][r] - m_new);  // exp(-inf - finite) = 0
+        m_run[qs][r] = m_new;
+        float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
+        float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
+        p_val[0][r] = p0;
+        p_val[1][r] = p1;
+        float row_sum = p0 + p1;
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          row_sum += __shfl_xor(row_sum, off);
+        l_run[qs][r] = l_run[qs][r] * alpha[r] + row_sum;
+      }
+
+      // rescale O
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) o_acc[qs][c][r] *= alpha[r];
+      }
+
+      // re-layout P (C layout) -> A layout via per-wave LDS tile
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          sm.p_tile[wave][qs][seg * 4 + r][kk * 16 + col] =
+              (short)f32_to_bf16(p_val[kk][r]);
+        }
       }
     }
+
     // same-wave LDS visibility: wait LDS ops, pin the scheduler
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_sched_barrier(0);
-    bf16x8 p_frag =
-        *reinterpret_cast<const bf16x8*>(&sm.p_tile[wave][col][seg * 8]);
 
-    // ---- O += P V : NC chunks of 16 output columns ----
+    // ---- O += P V : NC chunks of 16 output columns, per sub-tile ----
 #pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
-      bf16x8 v_frag =
-          *reinterpret_cast<const bf16x8*>(&sm.v_tile[c * 16 + col][seg * 8]);
-      o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
-                                                         o_acc[c], 0, 0, 0);
+    for (int qs = 0; qs < kQSub; ++qs) {
+      bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
+          &sm.p_tile[wave][qs][col][seg * 8]);
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
+        bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
+            &sm.v_tile[c * 16 + col][seg * 8]);
+        o_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            p_frag, v_frag, o_acc[qs][c], 0, 0, 0);
+      }
     }
   }
 
   // ---- epilogue: normalize, store O and LSE ----
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int q_row = q_row0 + seg * 4 + r;
-    if (q_row >= T) continue;
-    const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
-    const long out_base = qkv_off + (long)q_row * D;
+  for (int qs = 0; qs < kQSub; ++qs) {
 #pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      o[out_base + c * 16 + col] = (short)f32_to_bf16(o_acc[c][r] * inv_l);
-    }
-    if (col == 0) {
-      lse_out[bh * T + q_row] = m_run[r] + __logf(l_run[r]);
+    for (int r = 0; r < 4; ++r) {
+      const int q_row = q_row0 + 16 * qs + seg * 4 + r;
+      if (q_row >= T) continue;
+      const float inv_l = (l_run[qs][r] > 0.f) ? 1.f / l_run[qs][r] : 0.f;
+      const long out_base = qkv_off + (long)q_row * D;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        o[out_base + c * 16 + col] =
+            (short)f32_to_bf16(o_acc[qs][c][r] * inv_l);
+      }
+      if (col == 0) {
+        lse_out[bh * T + q_row] = m_run[qs][r] + __logf(l_run[qs][r]);
+      }
     }
   }
 }
@@ -325,7 +345,7 @@ template <int D>
 void launch_fmha_fwd(const torch::Tensor& q, const torch::Tensor& k,
                      const torch::Tensor& v, torch::Tensor& o,
                      torch::Tensor& lse, int T, long BH, float scale) {
-  dim3 grid((T + kQTile - 1) / kQTile, (unsigned)BH);
+  dim3 grid((unsigned)BH, (T + kQTile - 1) / kQTile);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(fmha_fwd_kernel<D>, grid, dim3(kBlockThreads),
                      sizeof(typename FmhaShapes<D>::Shared), stream,

@@ -192,6 +192,9 @@ def eval_on_val(val_loader, model, device):
 
 
 def main(cfg):
+    from .tuning import enable_tunableop
+
+    enable_tunableop()
     device = xdist.init_distributed()
     xdist.master_print(f"\n=== cfg ===\n{pprint.pformat(vars(cfg))}\n")
     xdist.master_print(f"device: {device}, world size: {xdist.get_world_size()}")
