@@ -5,7 +5,11 @@ plus a numerics check against the fp32 math reference.  Run on an MI355X:
     python benchmarks/bench_fmha.py
 """
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -21,7 +21,11 @@ def enable_tunableop():
     if _DONE or not torch.cuda.is_available():
         return
     _DONE = True
-    if os.environ.get("VITFSDP_TUNABLEOP", "1") == "0":
+    # opt-in only: TunableOp (both live tuning and read-only table load)
+    # was observed to crash the ROCm 7.2 / torch 2.10 stack on gfx950
+    # (core dump in the tuned bench run, gpurun_out/run3.log), so the
+    # default path never touches it.
+    if os.environ.get("VITFSDP_TUNABLEOP", "0") != "1":
         return
     csv = os.path.join(
         os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
