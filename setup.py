@@ -29,6 +29,7 @@ ext = CUDAExtension(
             "-O3",
             "-std=c++17",
             "--offload-arch=gfx950",
+            "-Rpass-analysis=kernel-resource-usage",
         ],
     },
 )
