@@ -33,6 +33,11 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     double scale);
+std::vector<torch::Tensor> fmha_fwd_qkv(torch::Tensor qkv, long num_heads,
+                                        double scale);
+torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
+                           torch::Tensor o, torch::Tensor lse,
+                           long num_heads, double scale);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -48,6 +53,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused softmax CE forward");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused softmax CE backward");
   m.def("fmha_fwd", &fmha_fwd, "flash attention forward (bf16, head_dim<=192)");
-  m.def("fmha_bwd", &fmha_bwd, "flash attention backward (recompute-based)");
+  m.def("fmha_bwd", &fmha_bwd, "flash attention backward (fused FA2-style)");
+  m.def("fmha_fwd_qkv", &fmha_fwd_qkv,
+        "flash attention forward on the fused [B,T,3,H,D] qkv projection "
+        "(zero-copy strided IO, O returned as [B,T,E])");
+  m.def("fmha_bwd_qkv", &fmha_bwd_qkv,
+        "flash attention backward producing the fused dqkv [B,T,3,H,D]");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
 }

@@ -98,11 +98,20 @@ struct FmhaShapes {
 // grid = (B*H, ceil(T/kQTile)): bh on x so a head's q-tiles share the
 // XCD whose L2 already holds its K/V (blockIdx linearization round-
 // robins x across XCDs; y strides keep bh%8 constant -> same XCD)
+// Strides (in elements, d innermost contiguous): the q/k/v pointers
+// share one stride set (they are views of the fused qkv projection
+// [B,T,3,H,D]); o has its own (it is written [B,T,H,D] = [B,T,E], so the
+// attention module needs no permute/contiguous copies on either side).
+struct QkvStrides {
+  long qb, qh, qt;  // q/k/v batch, head, token strides
+  long ob, oh, ot;  // o (and dO) strides
+};
+
 template <int D>
 __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse_out, int T, float scale) {
+    float* __restrict__ lse_out, int T, int H, QkvStrides st, float scale) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQSub = S::QSub;
@@ -119,7 +128,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
   // this wave's first q row, per sub-tile qs: q_row0 + 16*qs
   const int q_row0 = q_base + wave * (16 * kQSub);
 
-  const long qkv_off = bh * (long)T * D;
+  const int b_idx = (int)(bh / H), h_idx = (int)(bh % H);
+  const long qkv_off = (long)b_idx * st.qb + (long)h_idx * st.qh;
+  const long o_off = (long)b_idx * st.ob + (long)h_idx * st.oh;
 
   // ---- load Q fragments to registers: sub-tile qs, chunk kc, elem j ->
   // Q[q_row0 + 16*qs + col][seg*8 + j + 32*kc] (zero-padded beyond D/T) --
@@ -128,7 +139,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
   for (int qs = 0; qs < kQSub; ++qs) {
     const int q_row = q_row0 + 16 * qs + col;
     const bool valid = q_row < T;
-    const long base = qkv_off + (long)q_row * D;
+    const long base = qkv_off + (long)q_row * st.qt;
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
@@ -170,7 +181,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
         const int k_row = k_base + kr;
         if (k_row < T && dc < D) {
           val = *reinterpret_cast<const bf16x8*>(
-              &k[qkv_off + (long)k_row * D + dc]);
+              &k[qkv_off + (long)k_row * st.qt + dc]);
         }
         *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = val;
       }
@@ -186,7 +197,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
         bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
         if (k_row < T) {
           val = *reinterpret_cast<const bf16x8*>(
-              &v[qkv_off + (long)k_row * D + dc]);
+              &v[qkv_off + (long)k_row * st.qt + dc]);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) sm.v_tile[dc + j][kr] = val[j];
@@ -283,7 +294,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
       const int q_row = q_row0 + 16 * qs + seg * 4 + r;
       if (q_row >= T) continue;
       const float inv_l = (l_run[qs][r] > 0.f) ? 1.f / l_run[qs][r] : 0.f;
-      const long out_base = qkv_off + (long)q_row * D;
+      const long out_base = o_off + (long)q_row * st.ot;
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         o[out_base + c * 16 + col] =
@@ -314,7 +325,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, int T, float scale) {
+    short* __restrict__ dq, int T, int H, QkvStrides st, float scale) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQSub = S::QSub;
@@ -330,7 +341,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
   const long bh = blockIdx.x;
   const int q_base = blockIdx.y * kQTile;
   const int q_row0 = q_base + wave * (16 * kQSub);
-  const long qkv_off = bh * (long)T * D;
+  const int b_idx = (int)(bh / H), h_idx = (int)(bh % H);
+  const long qkv_off = (long)b_idx * st.qb + (long)h_idx * st.qh;
+  const long o_off = (long)b_idx * st.ob + (long)h_idx * st.oh;
 
   // Q and dO fragments in registers (A operands of S and dP)
   bf16x8 q_frag[kQSub][NKC], do_frag[kQSub][NKC];
@@ -339,7 +352,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
   for (int qs = 0; qs < kQSub; ++qs) {
     const int q_row = q_row0 + 16 * qs + col;
     const bool valid = q_row < T;
-    const long base = qkv_off + (long)q_row * D;
+    const long base = qkv_off + (long)q_row * st.qt;
+    const long dbase = o_off + (long)q_row * st.ot;
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
@@ -347,7 +361,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
       for (int j = 0; j < 8; ++j) {
         const bool ok = valid && d0 + j < D;
         q_frag[qs][kc][j] = ok ? q[base + d0 + j] : (short)0;
-        do_frag[qs][kc][j] = ok ? dout[base + d0 + j] : (short)0;
+        do_frag[qs][kc][j] = ok ? dout[dbase + d0 + j] : (short)0;
       }
     }
     // per-lane row stats for the 4 rows this lane's C fragments cover
@@ -382,9 +396,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
         bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
         if (k_row < T && dc < D) {
           kv = *reinterpret_cast<const bf16x8*>(
-              &k[qkv_off + (long)k_row * D + dc]);
+              &k[qkv_off + (long)k_row * st.qt + dc]);
           vv = *reinterpret_cast<const bf16x8*>(
-              &v[qkv_off + (long)k_row * D + dc]);
+              &v[qkv_off + (long)k_row * st.qt + dc]);
         }
         *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = kv;
         *reinterpret_cast<bf16x8*>(&sm.v_tile[kr][dc]) = vv;
@@ -455,7 +469,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
     for (int r = 0; r < 4; ++r) {
       const int q_row = q_row0 + 16 * qs + seg * 4 + r;
       if (q_row >= T) continue;
-      const long out_base = qkv_off + (long)q_row * D;
+      const long out_base = qkv_off + (long)q_row * st.qt;
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         dq[out_base + c * 16 + col] = (short)f32_to_bf16(dq_acc[qs][c][r]);
@@ -469,7 +483,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dk, short* __restrict__ dv, int T, float scale) {
+    short* __restrict__ dk, short* __restrict__ dv, int T, int H,
+    QkvStrides st, float scale) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQStream = 32;  // q rows streamed per iteration
@@ -484,14 +499,16 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
   const long bh = blockIdx.x;
   const int k_base = blockIdx.y * 64;     // workgroup's 64 k rows
   const int k_row0 = k_base + wave * 16;  // this wave's 16 k rows
-  const long qkv_off = bh * (long)T * D;
+  const int b_idx = (int)(bh / H), h_idx = (int)(bh % H);
+  const long qkv_off = (long)b_idx * st.qb + (long)h_idx * st.qh;
+  const long o_off = (long)b_idx * st.ob + (long)h_idx * st.oh;
 
   // K and V fragments in registers (A operands of S^T and dP^T)
   bf16x8 k_frag[NKC], v_frag[NKC];
   {
     const int k_row = k_row0 + col;
     const bool valid = k_row < T;
-    const long base = qkv_off + (long)k_row * D;
+    const long base = qkv_off + (long)k_row * st.qt;
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
@@ -528,9 +545,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
         bf16x8 dov = {0, 0, 0, 0, 0, 0, 0, 0};
         if (q_row < T && dc < D) {
           qv = *reinterpret_cast<const bf16x8*>(
-              &q[qkv_off + (long)q_row * D + dc]);
+              &q[qkv_off + (long)q_row * st.qt + dc]);
           dov = *reinterpret_cast<const bf16x8*>(
-              &dout[qkv_off + (long)q_row * D + dc]);
+              &dout[o_off + (long)q_row * st.ot + dc]);
         }
         *reinterpret_cast<bf16x8*>(&sm.q_tile[qr][dc]) = qv;
         *reinterpret_cast<bf16x8*>(&sm.do_tile[qr][dc]) = dov;
@@ -607,7 +624,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
   for (int r = 0; r < 4; ++r) {
     const int k_row = k_row0 + seg * 4 + r;
     if (k_row >= T) continue;
-    const long out_base = qkv_off + (long)k_row * D;
+    const long out_base = qkv_off + (long)k_row * st.qt;
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
       dk[out_base + c * 16 + col] = (short)f32_to_bf16(dk_acc[c][r]);
@@ -616,16 +633,20 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
   }
 }
 
-// Delta = rowsum(dO * O): one wave per row
+// Delta = rowsum(dO * O): one wave per (b,h,t) row; o/dO are strided
+// [B,T,H,D]-layout tensors, delta is contiguous [B,H,T]
 __global__ __launch_bounds__(64) void fmha_rowdot_kernel(
     const unsigned short* __restrict__ dout,
-    const unsigned short* __restrict__ o, float* __restrict__ delta, int D) {
-  const long row = blockIdx.x;
-  const unsigned short* dr = dout + row * D;
-  const unsigned short* orow = o + row * D;
+    const unsigned short* __restrict__ o, float* __restrict__ delta, int T,
+    int H, int D, long sob, long soh, long sot) {
+  const long row = blockIdx.x;  // linear over [B,H,T]
+  const int t = (int)(row % T);
+  const int h = (int)((row / T) % H);
+  const long b = row / ((long)T * H);
+  const long base = b * sob + (long)h * soh + (long)t * sot;
   float acc = 0.f;
   for (int i = threadIdx.x; i < D; i += 64)
-    acc += bf16_to_f32(dr[i]) * bf16_to_f32(orow[i]);
+    acc += bf16_to_f32(dout[base + i]) * bf16_to_f32(o[base + i]);
   acc = wave_all_sum(acc);
   if (threadIdx.x == 0) delta[row] = acc;
 }
@@ -693,67 +714,113 @@ __global__ __launch_bounds__(64) void mfma_probe_kernel(
 // explicit instantiations so -Rpass-analysis reports the real configs
 template __global__ void fmha_fwd_kernel<160>(const short*, const short*,
                                               const short*, short*, float*,
-                                              int, float);
+                                              int, int, QkvStrides, float);
 template __global__ void fmha_fwd_kernel<64>(const short*, const short*,
                                              const short*, short*, float*,
-                                             int, float);
+                                             int, int, QkvStrides, float);
 template __global__ void fmha_bwd_dq_kernel<160>(const short*, const short*,
                                                  const short*, const short*,
                                                  const float*, const float*,
-                                                 short*, int, float);
+                                                 short*, int, int, QkvStrides,
+                                                 float);
 template __global__ void fmha_bwd_dkv_kernel<160>(const short*, const short*,
                                                   const short*, const short*,
                                                   const float*, const float*,
-                                                  short*, short*, int, float);
+                                                  short*, short*, int, int,
+                                                  QkvStrides, float);
 #else
+struct FmhaArgs {
+  const short *q, *k, *v;
+  short* o;        // or dO for backward
+  int B, H, T;
+  QkvStrides st;
+  float scale;
+};
+
 template <int D>
-void launch_fmha_fwd(const torch::Tensor& q, const torch::Tensor& k,
-                     const torch::Tensor& v, torch::Tensor& o,
-                     torch::Tensor& lse, int T, long BH, float scale) {
-  dim3 grid((unsigned)BH,
-            (T + FmhaShapes<D>::QTile - 1) / FmhaShapes<D>::QTile);
+void launch_fmha_fwd(const FmhaArgs& a, torch::Tensor& lse) {
+  dim3 grid((unsigned)((long)a.B * a.H),
+            (a.T + FmhaShapes<D>::QTile - 1) / FmhaShapes<D>::QTile);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   hipLaunchKernelGGL(fmha_fwd_kernel<D>, grid, dim3(kBlockThreads),
-                     sizeof(typename FmhaShapes<D>::Shared), stream,
-                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                     (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                     lse.data_ptr<float>(), T, scale);
+                     sizeof(typename FmhaShapes<D>::Shared), stream, a.q, a.k,
+                     a.v, a.o, lse.data_ptr<float>(), a.T, a.H, a.st, a.scale);
   HIP_CHECK_LAST();
 }
 
 template <int D>
-void launch_fmha_bwd(const torch::Tensor& dout, const torch::Tensor& q,
-                     const torch::Tensor& k, const torch::Tensor& v,
-                     const torch::Tensor& lse, const torch::Tensor& delta,
-                     torch::Tensor& dq, torch::Tensor& dk, torch::Tensor& dv,
-                     int T, long BH, float scale) {
+void launch_fmha_bwd(const FmhaArgs& a, const torch::Tensor& lse,
+                     const torch::Tensor& delta, short* dq, short* dk,
+                     short* dv) {
   using S = FmhaShapes<D>;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  dim3 grid_dq((unsigned)BH, (T + S::QTile - 1) / S::QTile);
+  const unsigned BH = (unsigned)((long)a.B * a.H);
+  dim3 grid_dq(BH, (a.T + S::QTile - 1) / S::QTile);
   hipLaunchKernelGGL(fmha_bwd_dq_kernel<D>, grid_dq, dim3(kBlockThreads),
-                     sizeof(typename S::SharedDQ), stream,
-                     (const short*)q.data_ptr(),
-                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     (const short*)dout.data_ptr(), lse.data_ptr<float>(),
-                     delta.data_ptr<float>(), (short*)dq.data_ptr(), T,
-                     scale);
+                     sizeof(typename S::SharedDQ), stream, a.q, a.k, a.v,
+                     a.o /*dO*/, lse.data_ptr<float>(),
+                     delta.data_ptr<float>(), dq, a.T, a.H, a.st, a.scale);
   HIP_CHECK_LAST();
-  dim3 grid_dkv((unsigned)BH, (T + 63) / 64);
+  dim3 grid_dkv(BH, (a.T + 63) / 64);
   hipLaunchKernelGGL(fmha_bwd_dkv_kernel<D>, grid_dkv, dim3(kBlockThreads),
-                     sizeof(typename S::SharedDKV), stream,
-                     (const short*)q.data_ptr(),
-                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     (const short*)dout.data_ptr(), lse.data_ptr<float>(),
-                     delta.data_ptr<float>(), (short*)dk.data_ptr(),
-                     (short*)dv.data_ptr(), T, scale);
+                     sizeof(typename S::SharedDKV), stream, a.q, a.k, a.v,
+                     a.o /*dO*/, lse.data_ptr<float>(),
+                     delta.data_ptr<float>(), dk, dv, a.T, a.H, a.st,
+                     a.scale);
   HIP_CHECK_LAST();
 }
+
+#define VITFSDP_FMHA_DISPATCH(D_, expr)                               \
+  switch (D_) {                                                       \
+    case 16: { constexpr int kD = 16; expr; break; }                  \
+    case 32: { constexpr int kD = 32; expr; break; }                  \
+    case 48: { constexpr int kD = 48; expr; break; }                  \
+    case 64: { constexpr int kD = 64; expr; break; }                  \
+    case 80: { constexpr int kD = 80; expr; break; }                  \
+    case 96: { constexpr int kD = 96; expr; break; }                  \
+    case 112: { constexpr int kD = 112; expr; break; }                \
+    case 128: { constexpr int kD = 128; expr; break; }                \
+    case 144: { constexpr int kD = 144; expr; break; }                \
+    case 160: { constexpr int kD = 160; expr; break; }                \
+    case 176: { constexpr int kD = 176; expr; break; }                \
+    case 192: { constexpr int kD = 192; expr; break; }                \
+    default: TORCH_CHECK(false, "fmha: unsupported head_dim ", D_);   \
+  }
 
 #endif  // VITFSDP_KERNELS_ONLY
 
 }  // namespace
 
 #ifndef VITFSDP_KERNELS_ONLY
+namespace {
+
+QkvStrides contiguous_strides(int H, int T, int D) {
+  QkvStrides st;
+  st.qh = (long)T * D;
+  st.qb = (long)H * st.qh;
+  st.qt = D;
+  st.ob = st.qb;
+  st.oh = st.qh;
+  st.ot = st.qt;
+  return st;
+}
+
+torch::Tensor run_rowdot(const torch::Tensor& dout, const torch::Tensor& o,
+                         int B, int H, int T, int D, const QkvStrides& st) {
+  auto delta = torch::empty({B, H, T}, dout.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(fmha_rowdot_kernel, dim3((long)B * H * T), dim3(64), 0,
+                     stream, (const unsigned short*)dout.data_ptr(),
+                     (const unsigned short*)o.data_ptr(),
+                     delta.data_ptr<float>(), T, H, D, st.ob, st.oh, st.ot);
+  HIP_CHECK_LAST();
+  return delta;
+}
+
+}  // namespace
+
+// ---- [B,H,T,D] contiguous API (kernel unit tests, generic use) ----
+
 std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, double scale) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
@@ -766,24 +833,10 @@ std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
 
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
-  const long BH = (long)B * H;
-  const float s = (float)scale;
-  switch (D) {
-    case 16: launch_fmha_fwd<16>(q, k, v, o, lse, T, BH, s); break;
-    case 32: launch_fmha_fwd<32>(q, k, v, o, lse, T, BH, s); break;
-    case 48: launch_fmha_fwd<48>(q, k, v, o, lse, T, BH, s); break;
-    case 64: launch_fmha_fwd<64>(q, k, v, o, lse, T, BH, s); break;
-    case 80: launch_fmha_fwd<80>(q, k, v, o, lse, T, BH, s); break;
-    case 96: launch_fmha_fwd<96>(q, k, v, o, lse, T, BH, s); break;
-    case 112: launch_fmha_fwd<112>(q, k, v, o, lse, T, BH, s); break;
-    case 128: launch_fmha_fwd<128>(q, k, v, o, lse, T, BH, s); break;
-    case 144: launch_fmha_fwd<144>(q, k, v, o, lse, T, BH, s); break;
-    case 160: launch_fmha_fwd<160>(q, k, v, o, lse, T, BH, s); break;
-    case 176: launch_fmha_fwd<176>(q, k, v, o, lse, T, BH, s); break;
-    case 192: launch_fmha_fwd<192>(q, k, v, o, lse, T, BH, s); break;
-    default:
-      TORCH_CHECK(false, "fmha_fwd: unsupported head_dim ", D);
-  }
+  FmhaArgs a{(const short*)q.data_ptr(), (const short*)k.data_ptr(),
+             (const short*)v.data_ptr(), (short*)o.data_ptr(),
+             B, H, T, contiguous_strides(H, T, D), (float)scale};
+  VITFSDP_FMHA_DISPATCH(D, launch_fmha_fwd<kD>(a, lse));
   return {o, lse};
 }
 
@@ -793,14 +846,9 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
                                     double scale) {
   TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
   const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
+  const QkvStrides st = contiguous_strides(H, T, D);
+  auto delta = run_rowdot(dout, o, B, H, T, D, st);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-
-  auto delta = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
-  hipLaunchKernelGGL(fmha_rowdot_kernel, dim3(B * H * T), dim3(64), 0, stream,
-                     (const unsigned short*)dout.data_ptr(),
-                     (const unsigned short*)o.data_ptr(),
-                     delta.data_ptr<float>(), D);
-  HIP_CHECK_LAST();
 
   static const bool use_compose = [] {
     const char* e = getenv("VITFSDP_FMHA_BWD");
@@ -814,8 +862,8 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
     auto dp = at::matmul(dout, v.transpose(-2, -1));
     auto p = torch::empty_like(s_raw);
     auto ds = torch::empty_like(s_raw);
-    hipLaunchKernelGGL(fmha_dsoftmax_kernel, dim3(B * H * T), dim3(256), 0,
-                       stream, (const unsigned short*)s_raw.data_ptr(),
+    hipLaunchKernelGGL(fmha_dsoftmax_kernel, dim3((long)B * H * T), dim3(256),
+                       0, stream, (const unsigned short*)s_raw.data_ptr(),
                        (const unsigned short*)dp.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (unsigned short*)p.data_ptr(),
@@ -830,25 +878,70 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  const long BH = (long)B * H;
-  const float s = (float)scale;
-  switch (D) {
-    case 16: launch_fmha_bwd<16>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 32: launch_fmha_bwd<32>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 48: launch_fmha_bwd<48>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 64: launch_fmha_bwd<64>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 80: launch_fmha_bwd<80>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 96: launch_fmha_bwd<96>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 112: launch_fmha_bwd<112>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 128: launch_fmha_bwd<128>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 144: launch_fmha_bwd<144>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 160: launch_fmha_bwd<160>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 176: launch_fmha_bwd<176>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    case 192: launch_fmha_bwd<192>(dout, q, k, v, lse, delta, dq, dk, dv, T, BH, s); break;
-    default:
-      TORCH_CHECK(false, "fmha_bwd: unsupported head_dim ", D);
-  }
+  FmhaArgs a{(const short*)q.data_ptr(), (const short*)k.data_ptr(),
+             (const short*)v.data_ptr(), (short*)dout.data_ptr(),
+             B, H, T, st, (float)scale};
+  VITFSDP_FMHA_DISPATCH(
+      D, launch_fmha_bwd<kD>(a, lse, delta, (short*)dq.data_ptr(),
+                             (short*)dk.data_ptr(), (short*)dv.data_ptr()));
   return {dq, dk, dv};
+}
+
+// ---- fused-qkv API: zero-copy path used by the Attention module ----
+// qkv: [B, T, 3, H, D] contiguous (the fused projection output reshaped);
+// o comes back [B, T, H, D] (== [B, T, E]); dO arrives the same way.
+
+std::vector<torch::Tensor> fmha_fwd_qkv(torch::Tensor qkv, long num_heads,
+                                        double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 5 &&
+                  qkv.size(2) == 3 && qkv.size(3) == num_heads,
+              "fmha_fwd_qkv: expected contiguous [B, T, 3, H, D]");
+  TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16, "fmha: bf16 only");
+  const int B = qkv.size(0), T = qkv.size(1), H = qkv.size(3),
+            D = qkv.size(4);
+  TORCH_CHECK(D % 16 == 0 && D <= kMaxD,
+              "fmha: head_dim must be a multiple of 16 and <= ", kMaxD);
+  QkvStrides st;
+  st.qt = 3L * H * D;
+  st.qb = (long)T * st.qt;
+  st.qh = D;
+  st.ot = (long)H * D;
+  st.ob = (long)T * st.ot;
+  st.oh = D;
+
+  auto o = torch::empty({B, T, (long)H * D}, qkv.options());
+  auto lse = torch::empty({B, H, T}, qkv.options().dtype(torch::kFloat32));
+  const short* base = (const short*)qkv.data_ptr();
+  FmhaArgs a{base, base + (long)H * D, base + 2L * H * D,
+             (short*)o.data_ptr(), B, H, T, st, (float)scale};
+  VITFSDP_FMHA_DISPATCH(D, launch_fmha_fwd<kD>(a, lse));
+  return {o, lse};
+}
+
+torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
+                           torch::Tensor o, torch::Tensor lse,
+                           long num_heads, double scale) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous() && qkv.is_contiguous());
+  const int B = qkv.size(0), T = qkv.size(1), H = qkv.size(3),
+            D = qkv.size(4);
+  QkvStrides st;
+  st.qt = 3L * H * D;
+  st.qb = (long)T * st.qt;
+  st.qh = D;
+  st.ot = (long)H * D;
+  st.ob = (long)T * st.ot;
+  st.oh = D;
+  auto delta = run_rowdot(dout, o, B, H, T, D, st);
+
+  auto dqkv = torch::empty_like(qkv);
+  const short* base = (const short*)qkv.data_ptr();
+  short* dbase = (short*)dqkv.data_ptr();
+  FmhaArgs a{base, base + (long)H * D, base + 2L * H * D,
+             (short*)dout.data_ptr(), B, H, T, st, (float)scale};
+  VITFSDP_FMHA_DISPATCH(
+      D, launch_fmha_bwd<kD>(a, lse, delta, dbase, dbase + (long)H * D,
+                             dbase + 2L * H * D));
+  return dqkv;
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {

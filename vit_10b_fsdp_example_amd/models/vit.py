@@ -29,7 +29,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import dist as xdist
-from ..ops import LayerNorm, attention, cross_entropy  # noqa: F401
+from ..ops import LayerNorm, attention_qkv, cross_entropy  # noqa: F401
 
 
 def init_vit_weights(module, recursive=False):
@@ -101,16 +101,17 @@ class Attention(nn.Module):
 
     def forward(self, x):
         B, T, E = x.shape
+        # [B, T, 3, H, d] is a free view of the fused projection; the
+        # attention core reads q/k/v through strides and returns [B, T, E]
+        # directly (no permute/contiguous copies around the kernel)
         qkv = self.qkv(x).reshape(B, T, 3, self.num_heads, self.head_dim)
-        qkv = qkv.permute(2, 0, 3, 1, 4)  # [3, B, H, T, d]
-        q, k, v = qkv[0], qkv[1], qkv[2]
-        o = attention(
-            q, k, v,
+        o = attention_qkv(
+            qkv,
+            self.num_heads,
             scale=self.scale,
             dropout_p=self.attn_drop_p,
             training=self.training,
         )
-        o = o.transpose(1, 2).reshape(B, T, E)
         return self.proj_drop(self.proj(o))
 
 

@@ -1,6 +1,6 @@
 from ._extension import ext, has_ext, use_hip
 from .layernorm import LayerNorm, layer_norm
-from .attention import attention, math_attention
+from .attention import attention, attention_qkv, math_attention
 from .cross_entropy import CrossEntropyLoss, cross_entropy
 from .adamw import FusedAdamW
 from .multi_tensor import local_sqnorm, scale_
@@ -12,6 +12,7 @@ __all__ = [
     "LayerNorm",
     "layer_norm",
     "attention",
+    "attention_qkv",
     "math_attention",
     "CrossEntropyLoss",
     "cross_entropy",

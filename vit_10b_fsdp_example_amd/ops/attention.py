@@ -66,3 +66,50 @@ def attention(q, k, v, scale=None, dropout_p=0.0, training=False):
             q.contiguous(), k.contiguous(), v.contiguous(), scale
         )
     return math_attention(q, k, v, scale, dropout_p, training)
+
+
+class _FlashAttentionQkvFn(torch.autograd.Function):
+    """Zero-copy attention on the fused qkv projection [B,T,3,H,D]:
+    the kernels read q/k/v through strides and write O (and dqkv) in the
+    [B,T,E] layout the surrounding Linears use — no permute/contiguous
+    copies on either side of the attention core."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, scale):
+        o, lse = ext().fmha_fwd_qkv(qkv, num_heads, scale)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.num_heads = num_heads
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        dqkv = ext().fmha_bwd_qkv(
+            do.contiguous(), qkv, o, lse, ctx.num_heads, ctx.scale
+        )
+        return dqkv, None, None
+
+
+def attention_qkv(qkv, num_heads, scale=None, dropout_p=0.0, training=False):
+    """Attention on the fused qkv projection.
+
+    qkv: [B, T, 3, H, D] (a free reshape view of the qkv Linear output);
+    returns [B, T, H*D].  GPU bf16 uses the strided flash kernels; the
+    fallback path permutes to [B, H, T, D] and runs the math composition.
+    """
+    B, T, three, H, D = qkv.shape
+    assert three == 3 and H == num_heads
+    if scale is None:
+        scale = D ** -0.5
+    use_kernel = (
+        dropout_p == 0.0
+        and qkv.dtype in (torch.bfloat16,)
+        and use_hip(qkv)
+        and hasattr(ext(), "fmha_fwd_qkv")
+    )
+    if use_kernel:
+        return _FlashAttentionQkvFn.apply(qkv.contiguous(), num_heads, scale)
+    q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)
+    o = math_attention(q, k, v, scale, dropout_p, training)
+    return o.transpose(1, 2).reshape(B, T, H * D)

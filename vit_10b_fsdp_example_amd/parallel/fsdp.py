@@ -287,6 +287,13 @@ class FullyShardedDataParallel(nn.Module):
         self._materialize()
         _alloc_storage(self._full_grad, self._padded_numel * self._elem_bytes)
         self._full_grad.zero_()
+        # Pre-bind each view's .grad to its slice of the flat grad buffer:
+        # AccumulateGrad then adds the backward's gradients straight into
+        # the reduce-scatter payload (no separate copy pass).  If autograd
+        # ever rebinds .grad out-of-place, the post-accumulate hook
+        # detects it by data_ptr and falls back to an explicit copy.
+        for (_m, _n, shape, numel, off), view in zip(self._param_infos, self._views):
+            view.grad = self._full_grad.narrow(0, off, numel).view(shape)
         # prefetch the unit the backward will need next (previous in
         # forward order)
         if self._root_ref is not None:
@@ -304,7 +311,13 @@ class FullyShardedDataParallel(nn.Module):
             return
         _mod, _name, _shape, numel, off = self._param_infos[idx]
         _alloc_storage(self._full_grad, self._padded_numel * self._elem_bytes)
-        self._full_grad.narrow(0, off, numel).copy_(view.grad.reshape(-1))
+        expected_ptr = (
+            self._full_grad.data_ptr() + off * self._full_grad.element_size()
+        )
+        if view.grad.data_ptr() != expected_ptr:
+            # autograd rebound .grad out-of-place (or pre-binding was
+            # skipped): fall back to an explicit copy into the payload
+            self._full_grad.narrow(0, off, numel).copy_(view.grad.reshape(-1))
         view.grad = None
         self._grads_arrived += 1
         if self._grads_arrived == len(self._views):
