@@ -102,7 +102,8 @@ def test_ws2_ddp_matches_single_rank():
 def test_ws2_fsdp_grad_ckpt_and_noreshard():
     """ws=2 with grad ckpt off / reshard off still matches."""
     ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
-    for extra in (["--no_grad_ckpt"], ["--no_reshard_after_forward"]):
+    for extra in (["--no_grad_ckpt"], ["--no_reshard_after_forward"],
+                  ["--shard_on_cpu"]):
         results = run_multiprocess(_run_trajectory, world_size=2, args=(extra,))
         for l, g in results:
             np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)

@@ -258,6 +258,7 @@ def build_fsdp_vit_model(cfg, device, compute_dtype=torch.float32):
             flatten_parameters=cfg.flatten_parameters,
             compute_dtype=compute_dtype,
             device=device,
+            shard_on_cpu=cfg.shard_on_cpu,
         )
 
     model = FSDPViTModel(

@@ -68,6 +68,7 @@ class FullyShardedDataParallel(nn.Module):
         compute_dtype=torch.float32,
         device=None,
         prefetch=True,
+        shard_on_cpu=False,
     ):
         super().__init__()
         # `flatten_parameters` is accepted for CLI compatibility with the
@@ -80,6 +81,15 @@ class FullyShardedDataParallel(nn.Module):
         self.flatten_parameters = flatten_parameters
         self.compute_dtype = compute_dtype
         self.prefetch = prefetch
+        # Host-offload mode (reference --shard_on_cpu):
+        # beyond wrapping the module while it still lives on CPU, the fp32
+        # master shard and the AdamW state stay in pinned host memory;
+        # each gather stages the shard to the device with an async
+        # hipMemcpyAsync from pinned memory, and the reduced grad shard
+        # is copied back to the host for the CPU optimizer step.  Device
+        # memory then holds only the transient gathered params +
+        # activations (the 288 GB sizing lever for 60B-class models).
+        self.shard_on_cpu = shard_on_cpu
 
         self._comm = CommContext.get()
         ws, rank = self._comm.world_size, self._comm.rank
@@ -122,11 +132,18 @@ class FullyShardedDataParallel(nn.Module):
         flat = torch.zeros(self._padded_numel, dtype=torch.float32, device=src_device)
         for _, _, _, numel, off, p in infos:
             flat.narrow(0, off, numel).copy_(p.detach().reshape(-1).to(torch.float32))
-        shard = (
-            flat.narrow(0, rank * self._shard_numel, self._shard_numel)
-            .to(self.device)
-            .clone()
-        )
+        if shard_on_cpu:
+            shard = flat.narrow(
+                0, rank * self._shard_numel, self._shard_numel
+            ).clone()
+            if self.device.type == "cuda":
+                shard = shard.pin_memory()
+        else:
+            shard = (
+                flat.narrow(0, rank * self._shard_numel, self._shard_numel)
+                .to(self.device)
+                .clone()
+            )
         del flat
         self.flat_param = nn.Parameter(shard)
 
@@ -182,10 +199,15 @@ class FullyShardedDataParallel(nn.Module):
     # ------------------------------------------------------------------
 
     def _comm_shard(self):
-        """The shard in comm/compute dtype (cast from the fp32 master)."""
-        if self.flat_param.dtype == self.compute_dtype:
-            return self.flat_param.data
-        return self.flat_param.data.to(self.compute_dtype)
+        """The shard in comm/compute dtype on the compute device (cast
+        from the fp32 master; for --shard_on_cpu this is the pinned-host
+        -> device async staging copy)."""
+        data = self.flat_param.data
+        if data.device != self.device:
+            data = data.to(self.device, non_blocking=True)
+        if data.dtype != self.compute_dtype:
+            data = data.to(self.compute_dtype)
+        return data
 
     def _resident(self):
         return self._full_flat.untyped_storage().size() > 0
@@ -348,6 +370,8 @@ class FullyShardedDataParallel(nn.Module):
             g = out_shard.to(torch.float32)
             if self._comm.world_size > 1:
                 g.div_(self._comm.world_size)
+            if g.device != self.flat_param.device:
+                g = g.to(self.flat_param.device)  # host-offload: grad D2H
             if self.flat_param.grad is None:
                 self.flat_param.grad = g
             else:
