@@ -95,6 +95,17 @@ struct FmhaShapes {
   };
 };
 
+
+// Transposed LDS tiles ([d][k], k-stride 32+8) are written element-wise
+// (scalar b16 stores) during the tile transpose.  Without a swizzle,
+// lanes writing the same k for d-rows 8 apart land on ONE bank (row
+// stride 80 B == 0 mod 128 B): a ~20-way conflict.  XOR the k index
+// with bits >=3 of (d>>3): banks spread 4x, and the 16-byte-aligned
+// 8-element runs the b128 reads need stay contiguous.
+__device__ __forceinline__ int tr_swz(int d, int k) {
+  return k ^ (((d >> 3) & 3) << 3);
+}
+
 // grid = (B*H, ceil(T/kQTile)): bh on x so a head's q-tiles share the
 // XCD whose L2 already holds its K/V (blockIdx linearization round-
 // robins x across XCDs; y strides keep bh%8 constant -> same XCD)
@@ -200,7 +211,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
               &v[qkv_off + (long)k_row * st.qt + dc]);
         }
 #pragma unroll
-        for (int j = 0; j < 8; ++j) sm.v_tile[dc + j][kr] = val[j];
+        for (int j = 0; j < 8; ++j)
+          sm.v_tile[dc + j][tr_swz(dc + j, kr)] = val[j];
       }
     }
     __syncthreads();
@@ -268,7 +280,6 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
 
     // same-wave LDS visibility: wait LDS ops, pin the scheduler
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
 
     // ---- O += P V : NC chunks of 16 output columns, per sub-tile ----
 #pragma unroll
@@ -279,7 +290,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
       for (int c = 0; c < NC; ++c) {
         // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
         bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
-            &sm.v_tile[c * 16 + col][seg * 8]);
+            &sm.v_tile[c * 16 + col][tr_swz(c * 16 + col, seg * 8)]);
         o_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             p_frag, v_frag, o_acc[qs][c], 0, 0, 0);
       }
@@ -404,7 +415,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
         *reinterpret_cast<bf16x8*>(&sm.v_tile[kr][dc]) = vv;
         if (dc < D) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) sm.kt_tile[dc + j][kr] = kv[j];
+          for (int j = 0; j < 8; ++j)
+            sm.kt_tile[dc + j][tr_swz(dc + j, kr)] = kv[j];
         }
       }
     }
@@ -446,7 +458,6 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
 
     // dQ += dS K
 #pragma unroll
@@ -456,7 +467,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
-            &sm.kt_tile[c * 16 + col][seg * 8]);
+            &sm.kt_tile[c * 16 + col][tr_swz(c * 16 + col, seg * 8)]);
         dq_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             ds_frag, ktb, dq_acc[qs][c], 0, 0, 0);
       }
@@ -554,8 +565,8 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
         if (dc < D) {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            sm.qt_tile[dc + j][qr] = qv[j];
-            sm.dot_tile[dc + j][qr] = dov[j];
+            sm.qt_tile[dc + j][tr_swz(dc + j, qr)] = qv[j];
+            sm.dot_tile[dc + j][tr_swz(dc + j, qr)] = dov[j];
           }
         }
       }
@@ -601,7 +612,6 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
 
     bf16x8 pt_frag = *reinterpret_cast<const bf16x8*>(
         &sm.pt_tile[wave][col][seg * 8]);
@@ -609,10 +619,11 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
         &sm.dst_tile[wave][col][seg * 8]);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
+      const int swz = tr_swz(c * 16 + col, seg * 8);
       bf16x8 dotb = *reinterpret_cast<const bf16x8*>(
-          &sm.dot_tile[c * 16 + col][seg * 8]);
+          &sm.dot_tile[c * 16 + col][swz]);
       bf16x8 qtb = *reinterpret_cast<const bf16x8*>(
-          &sm.qt_tile[c * 16 + col][seg * 8]);
+          &sm.qt_tile[c * 16 + col][swz]);
       dv_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           pt_frag, dotb, dv_acc[c], 0, 0, 0);
       dk_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
