@@ -10,6 +10,16 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
                                          torch::Tensor rstd);
+std::vector<torch::Tensor> layernorm_add_fwd(torch::Tensor x,
+                                             torch::Tensor res,
+                                             torch::Tensor w, torch::Tensor b,
+                                             double eps);
+std::vector<torch::Tensor> layernorm_add_bwd(torch::Tensor dy,
+                                             torch::Tensor dsum,
+                                             torch::Tensor sum,
+                                             torch::Tensor w,
+                                             torch::Tensor mean,
+                                             torch::Tensor rstd);
 
 void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> grads,
@@ -44,6 +54,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (bf16, fp32 stats)");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward");
+  m.def("layernorm_add_fwd", &layernorm_add_fwd,
+        "fused residual-add + LayerNorm forward");
+  m.def("layernorm_add_bwd", &layernorm_add_bwd,
+        "fused residual-add + LayerNorm backward");
   m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (fp32)");
   m.def("multi_tensor_sqnorm", &multi_tensor_sqnorm,
         "sum of squares over tensor list");

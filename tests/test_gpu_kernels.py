@@ -255,3 +255,44 @@ def test_fmha_qkv_fused_path():
         qkv.grad.float().cpu().numpy(), qf.grad.cpu().numpy(),
         rtol=0.1, atol=0.06,
     )
+
+
+def test_fused_add_layernorm():
+    """Fused residual-add+LN matches the fp32 composition, forward and
+    backward (both inputs, gamma/beta), including the downstream use of
+    the summed output."""
+    from vit_10b_fsdp_example_amd.ops import fused_add_layer_norm
+
+    torch.manual_seed(4)
+    N, D = 512, 1024
+    x = torch.randn(4, N // 4, D, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    r = torch.randn_like(x, requires_grad=True)
+    w = torch.randn(D, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    b = torch.randn(D, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+
+    s, y = fused_add_layer_norm(x, r, w, b, 1e-6)
+    out = s * 0.5 + y * 2.0  # use both outputs downstream
+    dy = torch.randn_like(out)
+    out.backward(dy)
+
+    xf = x.detach().float().requires_grad_(True)
+    rf = r.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    sf = xf + rf
+    yf = torch.nn.functional.layer_norm(sf, (D,), wf, bf, 1e-6)
+    (sf * 0.5 + yf * 2.0).backward(dy.float())
+
+    np.testing.assert_allclose(s.detach().float().cpu(), sf.detach().cpu(),
+                               rtol=0.05, atol=0.03)
+    np.testing.assert_allclose(y.detach().float().cpu(), yf.detach().cpu(),
+                               rtol=0.05, atol=0.05)
+    np.testing.assert_allclose(x.grad.float().cpu(), xf.grad.cpu(),
+                               rtol=0.1, atol=0.1)
+    np.testing.assert_allclose(r.grad.float().cpu(), rf.grad.cpu(),
+                               rtol=0.1, atol=0.1)
+    np.testing.assert_allclose(w.grad.float().cpu(), wf.grad.cpu(),
+                               rtol=0.05, atol=0.02 * N ** 0.5)
+    np.testing.assert_allclose(b.grad.float().cpu(), bf.grad.cpu(),
+                               rtol=0.05, atol=0.02 * N ** 0.5)

@@ -64,6 +64,12 @@ def build_arg_parser():
         help="stop each epoch after N steps (0 = full epoch); for smoke "
         "tests and benchmarking",
     )
+    parser.add_argument(
+        "--profile", action="store_true", dest="profile",
+        help="profile a few early steps with torch.profiler (CPU+GPU "
+        "kernel timeline) and write a chrome trace + a top-kernel table "
+        "to --ckpt_dir",
+    )
     return parser
 
 

@@ -29,7 +29,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import dist as xdist
-from ..ops import LayerNorm, attention_qkv, cross_entropy  # noqa: F401
+from ..ops import LayerNorm, attention_qkv, cross_entropy, fused_add_layer_norm  # noqa: F401
 
 
 def init_vit_weights(module, recursive=False):
@@ -145,8 +145,12 @@ class Block(nn.Module):
         self.mlp = Mlp(dim, int(dim * mlp_ratio), drop=drop)
 
     def forward(self, x):
-        x = x + self.attn(self.norm1(x))
-        x = x + self.mlp(self.norm2(x))
+        attn_out = self.attn(self.norm1(x))
+        # fused residual add + norm2 (one kernel for x+attn_out and its LN)
+        x, normed = fused_add_layer_norm(
+            x, attn_out, self.norm2.weight, self.norm2.bias, self.norm2.eps
+        )
+        x = x + self.mlp(normed)
         return x
 
 

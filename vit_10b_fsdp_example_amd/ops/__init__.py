@@ -1,5 +1,5 @@
 from ._extension import ext, has_ext, use_hip
-from .layernorm import LayerNorm, layer_norm
+from .layernorm import LayerNorm, layer_norm, fused_add_layer_norm
 from .attention import attention, attention_qkv, math_attention
 from .cross_entropy import CrossEntropyLoss, cross_entropy
 from .adamw import FusedAdamW
@@ -11,6 +11,7 @@ __all__ = [
     "use_hip",
     "LayerNorm",
     "layer_norm",
+    "fused_add_layer_norm",
     "attention",
     "attention_qkv",
     "math_attention",

@@ -37,6 +37,43 @@ def layer_norm(x, weight, bias, eps=1e-6):
     return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
 
 
+class _FusedAddLayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps):
+        s, y, mean, rstd = ext().layernorm_add_fwd(x, res, weight, bias, eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return s, y
+
+    @staticmethod
+    def backward(ctx, dsum, dy):
+        s, weight, mean, rstd = ctx.saved_tensors
+        if dsum is None:
+            dsum = torch.zeros_like(dy)
+        dx1, dx2, dw, db = ext().layernorm_add_bwd(
+            dy.contiguous(), dsum.contiguous(), s, weight, mean, rstd
+        )
+        return dx1, dx2, dw, db, None
+
+
+def fused_add_layer_norm(x, res, weight, bias, eps=1e-6):
+    """Compute s = x + res and y = layer_norm(s) in one fused pass
+    (pre-LN residual pattern: norm2(x + attn_out)).  Returns (s, y).
+
+    The backward folds the residual gradient into the LN dx kernel, so
+    no separate elementwise add kernels run in either direction.
+    """
+    if (
+        x.dtype == torch.bfloat16
+        and x.shape[-1] % 8 == 0
+        and use_hip(x, res)
+    ):
+        return _FusedAddLayerNormFn.apply(
+            x.contiguous(), res.contiguous(), weight, bias, eps
+        )
+    s = x + res
+    return s, F.layer_norm(s, (s.shape[-1],), weight, bias, eps)
+
+
 class LayerNorm(nn.Module):
     """Drop-in LayerNorm module backed by the HIP kernel on GPU."""
 

@@ -112,6 +112,9 @@ def train(cfg):
 
     smoothed_loss = SmoothedValue(window_size=5)
     smoothed_time = SmoothedValue(window_size=5)
+    from .profiling import StepProfiler
+
+    profiler = StepProfiler(cfg, enabled=getattr(cfg, "profile", False))
     xdist.rendezvous("training begins")
     xdist.master_print("training begins")
     max_steps = getattr(cfg, "max_steps_per_epoch", 0)
@@ -146,6 +149,8 @@ def train(cfg):
             lr_scheduler.step()
             optimizer.zero_grad(set_to_none=True)
 
+            profiler.step()
+
             # 4. logging (deferred, event-gated)
             t_new = time.time()
             time_step_elapsed, time_step_b = t_new - time_step_b, t_new
@@ -158,6 +163,7 @@ def train(cfg):
                     args=(epoch, step, smoothed_loss, smoothed_time, loss, lr, device),
                 )
 
+        profiler.stop()
         xdist.drain_step_closures()
         time_epoch_elapsed = time.time() - time_epoch_b
         xdist.master_print(f"epoch {epoch} done ({time_epoch_elapsed:.2f} sec)")
