@@ -108,3 +108,13 @@ def test_ws2_fsdp_grad_ckpt_and_noreshard():
         for l, g in results:
             np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
             np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_ws4_fsdp_matches_single_rank():
+    """4-rank FSDP (gloo) reproduces the single-rank trajectory — more
+    ranks exercise padding (shard sizes) and collective ordering harder."""
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory, world_size=4, args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
