@@ -37,6 +37,7 @@ def parse():
     p.add_argument("--per_gpu_batch", type=int, default=PER_GPU_BATCH)
     p.add_argument("--shard_on_cpu", action="store_true")
     p.add_argument("--no_grad_ckpt", action="store_false", dest="grad_ckpt")
+    p.add_argument("--grad_ckpt_blocks", type=int, default=-1)
     return p.parse_args()
 
 
@@ -72,7 +73,9 @@ def main():
         "--num_classes", str(classes),
         "--batch_size", str(args.per_gpu_batch * world),
     ] + (["--shard_on_cpu"] if args.shard_on_cpu else [])
-      + ([] if args.grad_ckpt else ["--no_grad_ckpt"]))
+      + ([] if args.grad_ckpt else ["--no_grad_ckpt"])
+      + (["--grad_ckpt_blocks", str(args.grad_ckpt_blocks)]
+         if args.grad_ckpt_blocks >= 0 else []))
 
     torch.manual_seed(1234)
     t_build = time.time()
@@ -158,6 +161,7 @@ def main():
                 "seq_len": (img // patch) ** 2,
                 "parallelism": f"fsdp{world}" + ("+cpu_shard" if args.shard_on_cpu else ""),
                 "grad_ckpt": cfg.grad_ckpt,
+                "grad_ckpt_blocks": args.grad_ckpt_blocks,
                 "final_loss": float(loss.item()) if loss is not None else None,
             },
         }

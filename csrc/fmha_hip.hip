@@ -75,9 +75,9 @@ struct FmhaShapes {
   static constexpr int VStride = kKTile + 8;  // 40
   static constexpr int PStride = kKTile + 8;  // 40
   struct Shared {
-    short k_tile[kKTile][KStride];
-    short v_tile[D][VStride];            // transposed: [d][k]
-    short p_tile[4][QSub][16][PStride];  // per-wave P re-layout buffers
+    short k_tile[2][kKTile][KStride];       // double-buffered
+    short v_tile[2][D][VStride];            // transposed [d][k], dbuf
+    short p_tile[4][QSub][16][PStride];     // per-wave P re-layout buffers
   };
   struct SharedDQ {
     short k_tile[kKTile][KStride];   // row-major (S operand)
@@ -119,7 +119,9 @@ struct QkvStrides {
 };
 
 template <int D>
-__global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
+// min 3 waves/SIMD: the double-buffered staging otherwise lands at
+// 182 regs -> 2 waves; forcing 3 costs a few prologue spills at most
+__global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int T, int H, QkvStrides st, float scale) {
@@ -127,6 +129,9 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQSub = S::QSub;
   constexpr int kQTile = S::QTile;
+  // loads per thread per K/V tile (16-byte vectors)
+  constexpr int kKVecs = (kKTile * (DP / 8) + kBlockThreads - 1) / kBlockThreads;
+  constexpr int kVVecs = (kKTile * (D / 8) + kBlockThreads - 1) / kBlockThreads;
   HIP_DYNAMIC_SHARED(char, smem_raw)
   typename S::Shared& sm = *reinterpret_cast<typename S::Shared*>(smem_raw);
 
@@ -142,6 +147,22 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
   const int b_idx = (int)(bh / H), h_idx = (int)(bh % H);
   const long qkv_off = (long)b_idx * st.qb + (long)h_idx * st.qh;
   const long o_off = (long)b_idx * st.ob + (long)h_idx * st.oh;
+
+  // ---- hoisted staging coordinates (reused every K/V tile) ----
+  int k_kr[kKVecs], k_dc[kKVecs];
+#pragma unroll
+  for (int i = 0; i < kKVecs; ++i) {
+    const int idx = (int)threadIdx.x + i * kBlockThreads;
+    k_kr[i] = idx / (DP / 8);
+    k_dc[i] = (idx % (DP / 8)) * 8;
+  }
+  int v_kr[kVVecs], v_dc[kVVecs];
+#pragma unroll
+  for (int i = 0; i < kVVecs; ++i) {
+    const int idx = (int)threadIdx.x + i * kBlockThreads;
+    v_kr[i] = idx / (D / 8);
+    v_dc[i] = (idx % (D / 8)) * 8;
+  }
 
   // ---- load Q fragments to registers: sub-tile qs, chunk kc, elem j ->
   // Q[q_row0 + 16*qs + col][seg*8 + j + 32*kc] (zero-padded beyond D/T) --
@@ -176,46 +197,74 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     for (int c = 0; c < NC; ++c) o_acc[qs][c] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
+  // ---- staging helpers (double-buffered K/V, T14 split: the NEXT
+  // tile's global loads are issued before this tile's compute so HBM
+  // latency hides under the MFMA phases; the LDS writes land after) ----
+  // K and V staging share one register block: their live ranges are
+  // disjoint (K: issue -> write before PV; V: issue -> write after PV)
+  constexpr int kStageVecs = kKVecs > kVVecs ? kKVecs : kVVecs;
+  bf16x8 stage[kStageVecs];
+  auto issue_k_loads = [&](int k_base) {
+#pragma unroll
+    for (int i = 0; i < kKVecs; ++i) {
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int k_row = k_base + k_kr[i];
+      if (k_kr[i] < kKTile && k_row < T && k_dc[i] < D) {
+        val = *reinterpret_cast<const bf16x8*>(
+            &k[qkv_off + (long)k_row * st.qt + k_dc[i]]);
+      }
+      stage[i] = val;
+    }
+  };
+  auto write_k_tile = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < kKVecs; ++i) {
+      if (k_kr[i] < kKTile) {
+        *reinterpret_cast<bf16x8*>(&sm.k_tile[buf][k_kr[i]][k_dc[i]]) =
+            stage[i];
+      }
+    }
+  };
+  auto issue_v_loads = [&](int k_base) {
+#pragma unroll
+    for (int i = 0; i < kVVecs; ++i) {
+      bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int k_row = k_base + v_kr[i];
+      if (v_kr[i] < kKTile && k_row < T) {
+        val = *reinterpret_cast<const bf16x8*>(
+            &v[qkv_off + (long)k_row * st.qt + v_dc[i]]);
+      }
+      stage[i] = val;
+    }
+  };
+  auto write_v_tile = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < kVVecs; ++i) {
+      if (v_kr[i] < kKTile) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          sm.v_tile[buf][v_dc[i] + j][tr_swz(v_dc[i] + j, v_kr[i])] =
+              stage[i][j];
+        }
+      }
+    }
+  };
+
   const int n_ktiles = (T + kKTile - 1) / kKTile;
+  // prologue: stage tile 0 into buffer 0
+  issue_k_loads(0);
+  write_k_tile(0);
+  issue_v_loads(0);
+  write_v_tile(0);
+  __syncthreads();
+
+  int cur = 0;
   for (int kt = 0; kt < n_ktiles; ++kt) {
     const int k_base = kt * kKTile;
-    __syncthreads();  // previous iteration's LDS reads are done
+    const bool has_next = kt + 1 < n_ktiles;
 
-    // ---- cooperative K tile load: [32][D] -> sm.k_tile (16B vectors) ----
-    {
-      constexpr int vecs_per_row = DP / 8;
-      constexpr int total = kKTile * vecs_per_row;
-      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
-        const int kr = i / vecs_per_row;
-        const int dc = (i % vecs_per_row) * 8;
-        bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
-        const int k_row = k_base + kr;
-        if (k_row < T && dc < D) {
-          val = *reinterpret_cast<const bf16x8*>(
-              &k[qkv_off + (long)k_row * st.qt + dc]);
-        }
-        *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = val;
-      }
-    }
-    // ---- V tile load, transposed into [d][k] ----
-    {
-      constexpr int vecs_per_row = D / 8;
-      constexpr int total = kKTile * vecs_per_row;
-      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
-        const int kr = i / vecs_per_row;
-        const int dc = (i % vecs_per_row) * 8;
-        const int k_row = k_base + kr;
-        bf16x8 val = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (k_row < T) {
-          val = *reinterpret_cast<const bf16x8*>(
-              &v[qkv_off + (long)k_row * st.qt + dc]);
-        }
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          sm.v_tile[dc + j][tr_swz(dc + j, kr)] = val[j];
-      }
-    }
-    __syncthreads();
+    // issue next K loads; their latency hides under this tile's QK^T
+    if (has_next) issue_k_loads(k_base + kKTile);
 
     // ---- per q sub-tile: S = scale*QK^T, online softmax, stage P ----
 #pragma unroll
@@ -227,7 +276,7 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
         for (int kc = 0; kc < NKC; ++kc) {
           // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
           bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-              &sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
+              &sm.k_tile[cur][kk * 16 + col][kc * 32 + seg * 8]);
           s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[qs][kc], b_frag, s_frag[kk], 0, 0, 0);
         }
@@ -278,7 +327,14 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
       }
     }
 
-    // same-wave LDS visibility: wait LDS ops, pin the scheduler
+    // K(t+1) is in flight and QK^T(t) done: write it to the other buffer,
+    // then issue V(t+1) loads to hide under the PV phase
+    if (has_next) {
+      write_k_tile(cur ^ 1);
+      issue_v_loads(k_base + kKTile);
+    }
+
+    // same-wave LDS visibility for the staged P tiles
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // ---- O += P V : NC chunks of 16 output columns, per sub-tile ----
@@ -290,11 +346,16 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
       for (int c = 0; c < NC; ++c) {
         // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
         bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
-            &sm.v_tile[c * 16 + col][tr_swz(c * 16 + col, seg * 8)]);
+            &sm.v_tile[cur][c * 16 + col]
+                      [tr_swz(c * 16 + col, seg * 8)]);
         o_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             p_frag, v_frag, o_acc[qs][c], 0, 0, 0);
       }
     }
+
+    if (has_next) write_v_tile(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: normalize, store O and LSE ----
@@ -317,19 +378,6 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_fwd_kernel(
     }
   }
 }
-
-// ======================= fused flash backward =======================
-// Two kernels (FlashAttention-2 style), no batched GEMM library calls:
-//   dQ kernel — q-tile ownership (forward-like): recompute S and dP via
-//     MFMA, form dS in registers, accumulate dQ = dS K.  Each q row is
-//     owned by one block: plain bf16 stores, no atomics.
-//   dK/dV kernel — k-tile ownership: K,V live in registers as MFMA A
-//     fragments; q-tiles stream through LDS (row-major for the S^T/dP^T
-//     B operands + transposed copies for the dK/dV B operands); P^T and
-//     dS^T re-shaped through per-wave LDS tiles exactly like the
-//     forward's P.  Each k row owned by one block: no atomics.
-// dS = scale * P * (dP - Delta) carries the softmax scale, so
-// dQ = dS K and dK = dS^T Q need no extra scaling.
 
 template <int D>
 __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(

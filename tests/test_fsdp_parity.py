@@ -71,7 +71,8 @@ def test_ws1_fsdp_matches_ddp_modes():
     (fp32, dropout active so the grad-ckpt RNG path is exercised)."""
     ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"], dropout=True)
     for extra in ([], ["--no_grad_ckpt"], ["--no_reshard_after_forward"],
-                  ["--shard_on_cpu"], ["--flatten_parameters"]):
+                  ["--shard_on_cpu"], ["--flatten_parameters"],
+                  ["--grad_ckpt_blocks", "1"]):
         l, g = _single_rank_reference(extra, dropout=True)
         np.testing.assert_allclose(l, ref_l, rtol=1e-5, atol=1e-6, err_msg=str(extra))
         np.testing.assert_allclose(g, ref_g, rtol=1e-5, atol=1e-6, err_msg=str(extra))

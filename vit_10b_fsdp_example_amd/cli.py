@@ -65,6 +65,14 @@ def build_arg_parser():
         "tests and benchmarking",
     )
     parser.add_argument(
+        "--grad_ckpt_blocks", type=int, default=-1,
+        help="number of transformer blocks to wrap with activation "
+        "checkpointing (-1 = all when --grad_ckpt is on). With 288 GB of "
+        "HBM3E many configurations fit without recomputing every block; "
+        "the gradients are identical either way, only recompute work "
+        "changes.",
+    )
+    parser.add_argument(
         "--profile", action="store_true", dest="profile",
         help="profile a few early steps with torch.profiler (CPU+GPU "
         "kernel timeline) and write a chrome trace + a top-kernel table "

@@ -265,6 +265,21 @@ def build_fsdp_vit_model(cfg, device, compute_dtype=torch.float32):
             shard_on_cpu=cfg.shard_on_cpu,
         )
 
+    # --grad_ckpt_blocks N: checkpoint only the first N blocks (identical
+    # gradients; recompute is pure overhead wherever activations fit in
+    # the 288 GB of HBM3E).  -1 = every block, the reference behavior.
+    ckpt_limit = getattr(cfg, "grad_ckpt_blocks", -1)
+    ckpt_count = {"i": 0}
+
+    def grad_ckpt_wrap(module):
+        if not cfg.grad_ckpt:
+            return module
+        idx = ckpt_count["i"]
+        ckpt_count["i"] += 1
+        if 0 <= ckpt_limit <= idx:
+            return module
+        return checkpoint_module(module)
+
     model = FSDPViTModel(
         image_size=cfg.image_size,
         patch_size=cfg.patch_size,
@@ -276,7 +291,7 @@ def build_fsdp_vit_model(cfg, device, compute_dtype=torch.float32):
         mlp_dropout=cfg.mlp_dropout,
         att_dropout=cfg.att_dropout,
         num_classes=cfg.num_classes,
-        grad_ckpt_wrap=checkpoint_module if cfg.grad_ckpt else (lambda m: m),
+        grad_ckpt_wrap=grad_ckpt_wrap,
         fsdp_wrap=fsdp_wrap,
     )
     # root wrap without grad-ckpt (reference run_vit_training.py:197-199)
