@@ -587,39 +587,58 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
     dv_acc[c] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int n_qtiles = (T + kQStream - 1) / kQStream;
-  for (int qt = 0; qt < n_qtiles; ++qt) {
-    const int q_base = qt * kQStream;
-    __syncthreads();
-
-    // Q and dO tiles, row-major + transposed
-    {
-      constexpr int vecs_per_row = DP / 8;
-      constexpr int total = kQStream * vecs_per_row;
-      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
-        const int qr = i / vecs_per_row;
-        const int dc = (i % vecs_per_row) * 8;
-        const int q_row = q_base + qr;
-        bf16x8 qv = {0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 dov = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (q_row < T && dc < D) {
-          qv = *reinterpret_cast<const bf16x8*>(
-              &q[qkv_off + (long)q_row * st.qt + dc]);
-          dov = *reinterpret_cast<const bf16x8*>(
-              &dout[o_off + (long)q_row * st.ot + dc]);
-        }
-        *reinterpret_cast<bf16x8*>(&sm.q_tile[qr][dc]) = qv;
-        *reinterpret_cast<bf16x8*>(&sm.do_tile[qr][dc]) = dov;
-        if (dc < D) {
+  // hoisted staging coordinates (16B vectors of the q/dO tiles)
+  constexpr int kQVecs = (kQStream * (DP / 8) + kBlockThreads - 1) / kBlockThreads;
+  int s_qr[kQVecs], s_dc[kQVecs];
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            sm.qt_tile[dc + j][tr_swz(dc + j, qr)] = qv[j];
-            sm.dot_tile[dc + j][tr_swz(dc + j, qr)] = dov[j];
-          }
+  for (int i = 0; i < kQVecs; ++i) {
+    const int idx = (int)threadIdx.x + i * kBlockThreads;
+    s_qr[i] = idx / (DP / 8);
+    s_dc[i] = (idx % (DP / 8)) * 8;
+  }
+  bf16x8 q_st[kQVecs], do_st[kQVecs];
+  auto issue_qdo_loads = [&](int q_base) {
+#pragma unroll
+    for (int i = 0; i < kQVecs; ++i) {
+      bf16x8 qv = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf16x8 dov = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int q_row = q_base + s_qr[i];
+      if (s_qr[i] < kQStream && q_row < T && s_dc[i] < D) {
+        qv = *reinterpret_cast<const bf16x8*>(
+            &q[qkv_off + (long)q_row * st.qt + s_dc[i]]);
+        dov = *reinterpret_cast<const bf16x8*>(
+            &dout[o_off + (long)q_row * st.ot + s_dc[i]]);
+      }
+      q_st[i] = qv;
+      do_st[i] = dov;
+    }
+  };
+  auto write_qdo_tiles = [&]() {
+#pragma unroll
+    for (int i = 0; i < kQVecs; ++i) {
+      if (s_qr[i] >= kQStream) continue;
+      *reinterpret_cast<bf16x8*>(&sm.q_tile[s_qr[i]][s_dc[i]]) = q_st[i];
+      *reinterpret_cast<bf16x8*>(&sm.do_tile[s_qr[i]][s_dc[i]]) = do_st[i];
+      if (s_dc[i] < D) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          sm.qt_tile[s_dc[i] + j][tr_swz(s_dc[i] + j, s_qr[i])] = q_st[i][j];
+          sm.dot_tile[s_dc[i] + j][tr_swz(s_dc[i] + j, s_qr[i])] =
+              do_st[i][j];
         }
       }
     }
-    __syncthreads();
+  };
+
+  const int n_qtiles = (T + kQStream - 1) / kQStream;
+  issue_qdo_loads(0);
+  write_qdo_tiles();
+  __syncthreads();
+  for (int qt = 0; qt < n_qtiles; ++qt) {
+    const int q_base = qt * kQStream;
+    const bool has_next = qt + 1 < n_qtiles;
+    // next tile's HBM loads fly under this tile's MFMA phases (T14)
+    if (has_next) issue_qdo_loads(q_base + kQStream);
 
     // S^T and dP^T: [16 k][32 q]
     f32x4 st_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
@@ -676,6 +695,12 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
           pt_frag, dotb, dv_acc[c], 0, 0, 0);
       dk_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           dst_frag, qtb, dk_acc[c], 0, 0, 0);
+    }
+
+    if (has_next) {
+      __syncthreads();  // everyone is done reading tile qt
+      write_qdo_tiles();
+      __syncthreads();  // tile qt+1 fully staged
     }
   }
 
