@@ -101,7 +101,7 @@ def main():
         loss = loss_fn(model(x), y)
         loss.backward()
         if cfg.clip_grad_norm > 0:
-            model.clip_grad_norm_(cfg.clip_grad_norm)
+            model.clip_grad_norm_(cfg.clip_grad_norm, defer_scale=True)
         opt.step()
         sched.step()
         opt.zero_grad(set_to_none=True)

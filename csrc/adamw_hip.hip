@@ -39,7 +39,12 @@ struct TensorTable {
 
 __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
                                    float beta2, float eps, float wd,
-                                   float bias_c1, float bias_c2) {
+                                   float bias_c1, float bias_c2,
+                                   const float* __restrict__ grad_scale) {
+  // deferred gradient clipping: the clip coefficient (a device scalar
+  // from clip_grad_norm_(defer_scale=True)) scales g on the fly, which
+  // replaces a separate full read+write pass over every gradient shard
+  const float gs = grad_scale ? *grad_scale : 1.f;
   const float decay = 1.f - lr * wd;
   const float step_size = lr / bias_c1;
   const float inv_sqrt_c2 = rsqrtf(bias_c2);
@@ -55,7 +60,7 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
       float4 pv = p[i], gv = g[i], mv = m[i], vv = v[i];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        float pj = (&pv.x)[j], gj = (&gv.x)[j];
+        float pj = (&pv.x)[j], gj = (&gv.x)[j] * gs;
         float mj = beta1 * (&mv.x)[j] + (1.f - beta1) * gj;
         float vj = beta2 * (&vv.x)[j] + (1.f - beta2) * gj * gj;
         pj *= decay;
@@ -72,7 +77,7 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
     const long tail = tab.n[ti] & 3;
     if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
       const long i = tab.n[ti] - tail + threadIdx.x;
-      float pj = tab.p[ti][i], gj = tab.g[ti][i];
+      float pj = tab.p[ti][i], gj = tab.g[ti][i] * gs;
       float mj = beta1 * tab.m[ti][i] + (1.f - beta1) * gj;
       float vj = beta2 * tab.v[ti][i] + (1.f - beta2) * gj * gj;
       pj = pj * decay - step_size * mj / (sqrtf(vj) * inv_sqrt_c2 + eps);
@@ -140,7 +145,15 @@ void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> exp_avgs,
                  std::vector<torch::Tensor> exp_avg_sqs, double lr,
                  double beta1, double beta2, double eps, double weight_decay,
-                 double bias_c1, double bias_c2) {
+                 double bias_c1, double bias_c2,
+                 c10::optional<torch::Tensor> grad_scale) {
+  const float* gs_ptr = nullptr;
+  if (grad_scale.has_value()) {
+    TORCH_CHECK(grad_scale->is_cuda() &&
+                    grad_scale->scalar_type() == torch::kFloat32,
+                "grad_scale must be a CUDA fp32 scalar");
+    gs_ptr = grad_scale->data_ptr<float>();
+  }
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int n = (int)params.size();
   for (int base = 0; base < n; base += kMaxTensors) {
@@ -162,7 +175,7 @@ void fused_adamw(std::vector<torch::Tensor> params,
     hipLaunchKernelGGL(fused_adamw_kernel, dim3(grid_for(total)), dim3(kBlock),
                        0, stream, tab, (float)lr, (float)beta1, (float)beta2,
                        (float)eps, (float)weight_decay, (float)bias_c1,
-                       (float)bias_c2);
+                       (float)bias_c2, gs_ptr);
     HIP_CHECK_LAST();
   }
 }

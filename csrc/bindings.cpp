@@ -26,7 +26,8 @@ void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> exp_avgs,
                  std::vector<torch::Tensor> exp_avg_sqs, double lr,
                  double beta1, double beta2, double eps, double weight_decay,
-                 double bias_c1, double bias_c2);
+                 double bias_c1, double bias_c2,
+                 c10::optional<torch::Tensor> grad_scale);
 torch::Tensor multi_tensor_sqnorm(std::vector<torch::Tensor> tensors);
 void multi_tensor_scale(std::vector<torch::Tensor> tensors, double factor);
 void multi_tensor_scale_tensor(std::vector<torch::Tensor> tensors,

@@ -60,6 +60,16 @@ class FusedAdamW(torch.optim.Optimizer):
             bias_c1 = 1.0 - beta1 ** step_t
             bias_c2 = 1.0 - beta2 ** step_t
 
+            # deferred gradient clipping (clip_grad_norm_(defer_scale=True)
+            # leaves a device-scalar coefficient on the params instead of
+            # running a separate full scale pass over the grads)
+            grad_scale = None
+            for p in params:
+                s = getattr(p, "_deferred_grad_scale", None)
+                if s is not None:
+                    grad_scale = s if grad_scale is None else grad_scale
+                    p._deferred_grad_scale = None
+
             if params[0].is_cuda and ext() is not None:
                 ext().fused_adamw(
                     params,
@@ -73,8 +83,11 @@ class FusedAdamW(torch.optim.Optimizer):
                     group["weight_decay"],
                     bias_c1,
                     bias_c2,
+                    grad_scale,
                 )
             else:
+                if grad_scale is not None:
+                    grads = torch._foreach_mul(grads, float(grad_scale))
                 self._foreach_step(
                     params, grads, exp_avgs, exp_avg_sqs,
                     group["lr"], beta1, beta2, group["eps"],

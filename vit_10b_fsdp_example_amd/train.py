@@ -137,7 +137,10 @@ def train(cfg):
                 # shards partition the full gradient, so one scalar
                 # all-reduce of the local sq-norms is exact
                 if cfg.clip_grad_norm > 0:
-                    model.clip_grad_norm_(cfg.clip_grad_norm)
+                    # defer_scale: FusedAdamW folds the clip coefficient
+                    # into its gradient read (no separate scale pass)
+                    model.clip_grad_norm_(cfg.clip_grad_norm,
+                                          defer_scale=True)
             else:
                 # plain-DDP baseline (reference run_vit_training.py:271-275)
                 xdist.reduce_gradients(optimizer)
