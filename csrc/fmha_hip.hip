@@ -380,7 +380,7 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
 }
 
 template <int D>
-__global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
+__global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -438,37 +438,55 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
 #pragma unroll
     for (int c = 0; c < NC; ++c) dq_acc[qs][c] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int n_ktiles = (T + kKTile - 1) / kKTile;
-  for (int kt = 0; kt < n_ktiles; ++kt) {
-    const int k_base = kt * kKTile;
-    __syncthreads();
-
-    // K and V row-major tiles + transposed K
-    {
-      constexpr int vecs_per_row = DP / 8;
-      constexpr int total = kKTile * vecs_per_row;
-      for (int i = threadIdx.x; i < total; i += kBlockThreads) {
-        const int kr = i / vecs_per_row;
-        const int dc = (i % vecs_per_row) * 8;
-        const int k_row = k_base + kr;
-        bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
-        if (k_row < T && dc < D) {
-          kv = *reinterpret_cast<const bf16x8*>(
-              &k[qkv_off + (long)k_row * st.qt + dc]);
-          vv = *reinterpret_cast<const bf16x8*>(
-              &v[qkv_off + (long)k_row * st.qt + dc]);
-        }
-        *reinterpret_cast<bf16x8*>(&sm.k_tile[kr][dc]) = kv;
-        *reinterpret_cast<bf16x8*>(&sm.v_tile[kr][dc]) = vv;
-        if (dc < D) {
+  // hoisted staging coordinates + register staging for the K/V tiles
+  constexpr int kLVecs = (kKTile * (DP / 8) + kBlockThreads - 1) / kBlockThreads;
+  int l_kr[kLVecs], l_dc[kLVecs];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            sm.kt_tile[dc + j][tr_swz(dc + j, kr)] = kv[j];
-        }
+  for (int i = 0; i < kLVecs; ++i) {
+    const int idx = (int)threadIdx.x + i * kBlockThreads;
+    l_kr[i] = idx / (DP / 8);
+    l_dc[i] = (idx % (DP / 8)) * 8;
+  }
+  bf16x8 kst[kLVecs], vst[kLVecs];
+  auto issue_kv_loads = [&](int k_base) {
+#pragma unroll
+    for (int i = 0; i < kLVecs; ++i) {
+      bf16x8 kv = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf16x8 vv = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int k_row = k_base + l_kr[i];
+      if (l_kr[i] < kKTile && k_row < T && l_dc[i] < D) {
+        kv = *reinterpret_cast<const bf16x8*>(
+            &k[qkv_off + (long)k_row * st.qt + l_dc[i]]);
+        vv = *reinterpret_cast<const bf16x8*>(
+            &v[qkv_off + (long)k_row * st.qt + l_dc[i]]);
+      }
+      kst[i] = kv;
+      vst[i] = vv;
+    }
+  };
+  auto write_kv_tiles = [&]() {
+#pragma unroll
+    for (int i = 0; i < kLVecs; ++i) {
+      if (l_kr[i] >= kKTile) continue;
+      *reinterpret_cast<bf16x8*>(&sm.k_tile[l_kr[i]][l_dc[i]]) = kst[i];
+      *reinterpret_cast<bf16x8*>(&sm.v_tile[l_kr[i]][l_dc[i]]) = vst[i];
+      if (l_dc[i] < D) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          sm.kt_tile[l_dc[i] + j][tr_swz(l_dc[i] + j, l_kr[i])] = kst[i][j];
       }
     }
-    __syncthreads();
+  };
+
+  const int n_ktiles = (T + kKTile - 1) / kKTile;
+  issue_kv_loads(0);
+  write_kv_tiles();
+  __syncthreads();
+  for (int kt = 0; kt < n_ktiles; ++kt) {
+    const int k_base = kt * kKTile;
+    const bool has_next = kt + 1 < n_ktiles;
+    // next tile's HBM loads fly under this tile's MFMA phases (T14)
+    if (has_next) issue_kv_loads(k_base + kKTile);
 
 #pragma unroll
     for (int qs = 0; qs < kQSub; ++qs) {
@@ -519,6 +537,12 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dq_kernel(
         dq_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             ds_frag, ktb, dq_acc[qs][c], 0, 0, 0);
       }
+    }
+
+    if (has_next) {
+      __syncthreads();  // all reads of tile kt done
+      write_kv_tiles();
+      __syncthreads();  // tile kt+1 staged
     }
   }
 
