@@ -74,9 +74,8 @@ struct FmhaShapes {
   static constexpr int VStride = kKTile + 8;  // 40
   static constexpr int PStride = kKTile + 8;  // 40
   struct Shared {
-    short k_tile[2][kKTile][KStride];       // double-buffered
-    short v_tile[2][D][VStride];            // transposed [d][k], dbuf
-    short p_tile[4][QSub][16][PStride];     // per-wave P re-layout buffers
+    short k_tile[2][kKTile][KStride];  // double-buffered
+    short v_tile[2][D][VStride];       // transposed [d][k], dbuf
   };
   struct SharedDQ {
     short k_tile[kKTile][KStride];   // row-major (S operand)
@@ -182,16 +181,13 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
     }
   }
 
-  // ---- online softmax state (per lane: 4 q rows per sub-tile) ----
-  float m_run[kQSub][4], l_run[kQSub][4];
+  // ---- online softmax state: ONE q row per lane (q = col) ----
+  float m_run[kQSub], l_run[kQSub];
   f32x4 o_acc[kQSub][NC];
 #pragma unroll
   for (int qs = 0; qs < kQSub; ++qs) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      m_run[qs][r] = -INFINITY;
-      l_run[qs][r] = 0.f;
-    }
+    m_run[qs] = -INFINITY;
+    l_run[qs] = 0.f;
 #pragma unroll
     for (int c = 0; c < NC; ++c) o_acc[qs][c] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
@@ -265,82 +261,118 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
     // issue next K loads; their latency hides under this tile's QK^T
     if (has_next) issue_k_loads(k_base + kKTile);
 
-    // ---- per q sub-tile: S = scale*QK^T, online softmax, stage P ----
+    // ---- per q sub-tile: S^T = scale*(K Q^T), in-register softmax ----
+    // Swapped operands (guide §B attn): mfma(A=K, B=Q) puts the scores
+    // in C layout rows = k, cols = q — each lane holds the 8 scores of
+    // ONE q row (q = col), so the row reduce is 7 in-register ops plus
+    // two cross-seg shuffles, the softmax state is two scalars per
+    // lane, and P returns to the MFMA A layout with four packed
+    // shuffles instead of an LDS round-trip + lgkmcnt(0) drain.
+    // Q's register fragments serve both orders: its A-layout fragment
+    // (lane: Q[col][seg*8+j]) IS its B-layout fragment for the swap.
 #pragma unroll
     for (int qs = 0; qs < kQSub; ++qs) {
-      f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+      f32x4 st_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
         for (int kc = 0; kc < NKC; ++kc) {
-          // B fragment: B[seg*8+j][col] = K[kk*16+col][kc*32+seg*8+j]
-          bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+          // A fragment: A[col][seg*8+j] = K[kk*16+col][kc*32+seg*8+j]
+          bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
               &sm.k_tile[cur][kk * 16 + col][kc * 32 + seg * 8]);
-          s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              q_frag[qs][kc], b_frag, s_frag[kk], 0, 0, 0);
+          st_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, q_frag[qs][kc], st_frag[kk], 0, 0, 0);
         }
       }
 
-      float p_val[2][4];
-      float alpha[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float s0 = s_frag[0][r] * scale;
-        float s1 = s_frag[1][r] * scale;
-        if (k_base + col >= T) s0 = -INFINITY;
-        if (k_base + 16 + col >= T) s1 = -INFINITY;
-        float m_tile = fmaxf(s0, s1);
-        // row max across the 16-lane column group
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-          m_tile = fmaxf(m_tile, __shfl_xor(m_tile, off));
-        const float m_new = fmaxf(m_run[qs][r], m_tile);
-        alpha[r] = __expf(m_run[qs][r] - m_new);  // exp(-inf - finite) = 0
-        m_run[qs][r] = m_new;
-        float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
-        float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
-        p_val[0][r] = p0;
-        p_val[1][r] = p1;
-        float row_sum = p0 + p1;
-#pragma unroll
-        for (int off = 1; off < 16; off <<= 1)
-          row_sum += __shfl_xor(row_sum, off);
-        l_run[qs][r] = l_run[qs][r] * alpha[r] + row_sum;
-      }
-
-      // rescale O
-#pragma unroll
-      for (int c = 0; c < NC; ++c) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[qs][c][r] *= alpha[r];
-      }
-
-      // re-layout P (C layout) -> A layout via per-wave LDS tile
+      // lane's 8 scores: q = col, k = k_base + 16*kk + seg*4 + r
+      float sc[2][4];
+      float my_max = -INFINITY;
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          sm.p_tile[wave][qs][seg * 4 + r][kk * 16 + col] =
-              (short)f32_to_bf16(p_val[kk][r]);
+          float sv = st_frag[kk][r] * scale;
+          if (k_base + 16 * kk + seg * 4 + r >= T) sv = -INFINITY;
+          sc[kk][r] = sv;
+          my_max = fmaxf(my_max, sv);
         }
       }
-    }
+      my_max = fmaxf(my_max, __shfl_xor(my_max, 16));
+      my_max = fmaxf(my_max, __shfl_xor(my_max, 32));
+      const float m_new = fmaxf(m_run[qs], my_max);
+      const float alpha = __expf(m_run[qs] - m_new);
+      m_run[qs] = m_new;
 
-    // K(t+1) is in flight and QK^T(t) done: write it to the other buffer,
-    // then issue V(t+1) loads to hide under the PV phase
-    if (has_next) {
-      write_k_tile(cur ^ 1);
-      issue_v_loads(k_base + kKTile);
-    }
-
-    // same-wave LDS visibility for the staged P tiles
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-    // ---- O += P V : NC chunks of 16 output columns, per sub-tile ----
+      float p[2][4];
+      float row_sum = 0.f;
 #pragma unroll
-    for (int qs = 0; qs < kQSub; ++qs) {
-      bf16x8 p_frag = *reinterpret_cast<const bf16x8*>(
-          &sm.p_tile[wave][qs][col][seg * 8]);
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float pv =
+              (sc[kk][r] == -INFINITY) ? 0.f : __expf(sc[kk][r] - m_new);
+          p[kk][r] = pv;
+          row_sum += pv;
+        }
+      }
+      row_sum += __shfl_xor(row_sum, 16);
+      row_sum += __shfl_xor(row_sum, 32);
+      l_run[qs] = l_run[qs] * alpha + row_sum;
+
+      // pack P to bf16 pairs (consecutive k): pk[kk][rr] holds
+      // (k = 16*kk + seg*4 + 2*rr, +1) for q = col
+      unsigned pk[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+          pk[kk][rr] = (unsigned)f32_to_bf16(p[kk][2 * rr]) |
+                       ((unsigned)f32_to_bf16(p[kk][2 * rr + 1]) << 16);
+        }
+      }
+      // redistribute to the MFMA A layout: target elem pair j2 wants
+      // k = seg*8 + 2*j2 of its own q=col, which lives in lane
+      // col + 16*((seg&1)*2 + (j2>>1)), slot (kk=seg>>1, rr=j2&1)
+      const int kk_s = seg >> 1;
+      unsigned pa_u[4];
+#pragma unroll
+      for (int j2 = 0; j2 < 4; ++j2) {
+        const int src = col + 16 * ((seg & 1) * 2 + (j2 >> 1));
+        const unsigned v = (j2 & 1) ? (kk_s ? pk[1][1] : pk[0][1])
+                                    : (kk_s ? pk[1][0] : pk[0][0]);
+        pa_u[j2] = (unsigned)__shfl((int)v, src);
+      }
+      bf16x8 pa;
+      {
+        union {
+          unsigned u[4];
+          bf16x8 v;
+        } cvt;
+        cvt.u[0] = pa_u[0];
+        cvt.u[1] = pa_u[1];
+        cvt.u[2] = pa_u[2];
+        cvt.u[3] = pa_u[3];
+        pa = cvt.v;
+      }
+
+      // rescale O: its C-layout rows are q = seg*4 + r, whose alpha
+      // lives in lane (seg*4 + r) of the column group
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float alpha_o = __shfl(alpha, seg * 4 + r);
+#pragma unroll
+        for (int c = 0; c < NC; ++c) o_acc[qs][c][r] *= alpha_o;
+      }
+
+      // K(t+1) in flight and QK^T(t) done for the LAST sub-tile: write
+      // it, then issue V(t+1) loads to hide under PV
+      if (qs == kQSub - 1 && has_next) {
+        write_k_tile(cur ^ 1);
+        issue_v_loads(k_base + kKTile);
+      }
+
+      // ---- O += P V : NC chunks of 16 output columns ----
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
@@ -348,7 +380,7 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
             &sm.v_tile[cur][c * 16 + col]
                       [tr_swz(c * 16 + col, seg * 8)]);
         o_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            p_frag, v_frag, o_acc[qs][c], 0, 0, 0);
+            pa, v_frag, o_acc[qs][c], 0, 0, 0);
       }
     }
 
@@ -360,19 +392,25 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
   // ---- epilogue: normalize, store O and LSE ----
 #pragma unroll
   for (int qs = 0; qs < kQSub; ++qs) {
+    // this lane's own row stats (q = col) feed the LSE store; the O
+    // rows it holds (q = seg*4 + r) read their l from the owning lane
+    if (seg == 0) {
+      const int q_row = q_row0 + 16 * qs + col;
+      if (q_row < T) {
+        lse_out[bh * T + q_row] = m_run[qs] + __logf(l_run[qs]);
+      }
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
+      const float l_row = __shfl(l_run[qs], seg * 4 + r);
       const int q_row = q_row0 + 16 * qs + seg * 4 + r;
       if (q_row >= T) continue;
-      const float inv_l = (l_run[qs][r] > 0.f) ? 1.f / l_run[qs][r] : 0.f;
+      const float inv_l = (l_row > 0.f) ? 1.f / l_row : 0.f;
       const long out_base = o_off + (long)q_row * st.ot;
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         o[out_base + c * 16 + col] =
             (short)f32_to_bf16(o_acc[qs][c][r] * inv_l);
-      }
-      if (col == 0) {
-        lse_out[bh * T + q_row] = m_run[qs][r] + __logf(l_run[qs][r]);
       }
     }
   }
