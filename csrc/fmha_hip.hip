@@ -334,15 +334,17 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
       }
       // redistribute to the MFMA A layout: target elem pair j2 wants
       // k = seg*8 + 2*j2 of its own q=col, which lives in lane
-      // col + 16*((seg&1)*2 + (j2>>1)), slot (kk=seg>>1, rr=j2&1)
+      // col + 16*((seg&1)*2 + (j2>>1)), slot (kk=seg>>1, rr=j2&1).
+      // A source lane serves targets from BOTH kk halves, so shuffle
+      // both kk slots and let each target select by its own seg.
       const int kk_s = seg >> 1;
       unsigned pa_u[4];
 #pragma unroll
       for (int j2 = 0; j2 < 4; ++j2) {
         const int src = col + 16 * ((seg & 1) * 2 + (j2 >> 1));
-        const unsigned v = (j2 & 1) ? (kk_s ? pk[1][1] : pk[0][1])
-                                    : (kk_s ? pk[1][0] : pk[0][0]);
-        pa_u[j2] = (unsigned)__shfl((int)v, src);
+        const unsigned lo = (unsigned)__shfl((int)pk[0][j2 & 1], src);
+        const unsigned hi = (unsigned)__shfl((int)pk[1][j2 & 1], src);
+        pa_u[j2] = kk_s ? hi : lo;
       }
       bf16x8 pa;
       {
