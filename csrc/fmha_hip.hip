@@ -79,10 +79,9 @@ struct FmhaShapes {
     short v_tile[2][D][VStride];       // transposed [d][k], dbuf
   };
   struct SharedDQ {
-    short k_tile[kKTile][KStride];   // row-major (S operand)
-    short v_tile[kKTile][KStride];   // row-major (dP operand)
-    short kt_tile[D][VStride];       // transposed K (dQ operand)
-    short ds_tile[4][QSub][16][PStride];  // dS re-layout per wave
+    short k_tile[kKTile][KStride];  // row-major (S^T operand)
+    short v_tile[kKTile][KStride];  // row-major (dP^T operand)
+    short kt_tile[D][VStride];      // transposed K (dQ operand)
   };
   struct SharedDKV {
     short q_tile[32][KStride];    // row-major (S^T operand)
@@ -444,9 +443,10 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
   const long qkv_off = (long)b_idx * st.qb + (long)h_idx * st.qh;
   const long o_off = (long)b_idx * st.ob + (long)h_idx * st.oh;
 
-  // Q and dO fragments in registers (A operands of S and dP)
+  // Q and dO register fragments: the same registers serve as the B
+  // operands of the swapped S^T / dP^T products (see the forward)
   bf16x8 q_frag[kQSub][NKC], do_frag[kQSub][NKC];
-  float lse_r[kQSub][4], delta_r[kQSub][4];
+  float lse_q[kQSub], delta_q[kQSub];
 #pragma unroll
   for (int qs = 0; qs < kQSub; ++qs) {
     const int q_row = q_row0 + 16 * qs + col;
@@ -463,13 +463,9 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
         do_frag[qs][kc][j] = ok ? dout[dbase + d0 + j] : (short)0;
       }
     }
-    // per-lane row stats for the 4 rows this lane's C fragments cover
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int row = q_row0 + 16 * qs + seg * 4 + r;
-      lse_r[qs][r] = (row < T) ? lse[bh * T + row] : 0.f;
-      delta_r[qs][r] = (row < T) ? delta[bh * T + row] : 0.f;
-    }
+    // per-lane row stats: this lane's q row is q = col (swapped layout)
+    lse_q[qs] = valid ? lse[bh * T + q_row] : 0.f;
+    delta_q[qs] = valid ? delta[bh * T + q_row] : 0.f;
   }
 
   f32x4 dq_acc[kQSub][NC];
@@ -530,8 +526,10 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
 
 #pragma unroll
     for (int qs = 0; qs < kQSub; ++qs) {
-      f32x4 s_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-      f32x4 dp_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+      // swapped: S^T = K Q^T, dP^T = V dO^T — C rows = k, cols = q, so
+      // each lane owns one q row (q = col) like the forward
+      f32x4 st_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+      f32x4 dpt_frag[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
@@ -540,36 +538,60 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
               &sm.k_tile[kk * 16 + col][kc * 32 + seg * 8]);
           bf16x8 vb = *reinterpret_cast<const bf16x8*>(
               &sm.v_tile[kk * 16 + col][kc * 32 + seg * 8]);
-          s_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              q_frag[qs][kc], kb, s_frag[kk], 0, 0, 0);
-          dp_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              do_frag[qs][kc], vb, dp_frag[kk], 0, 0, 0);
+          st_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              kb, q_frag[qs][kc], st_frag[kk], 0, 0, 0);
+          dpt_frag[kk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              vb, do_frag[qs][kc], dpt_frag[kk], 0, 0, 0);
         }
       }
-      // dS = scale * exp(scale*S - lse) * (dP - Delta), masked beyond T
+
+      // dS = scale * exp(scale*S - lse) * (dP - Delta), packed to bf16
+      // pairs along k (k = k_base + 16*kk + seg*4 + r for q = col)
+      unsigned pk[2][2];
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
-        const bool k_ok = (k_base + kk * 16 + col) < T;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float ds = 0.f;
-          if (k_ok) {
-            const float p =
-                __expf(s_frag[kk][r] * scale - lse_r[qs][r]);
-            ds = scale * p * (dp_frag[kk][r] - delta_r[qs][r]);
+        for (int rr = 0; rr < 2; ++rr) {
+          unsigned packed = 0;
+#pragma unroll
+          for (int h = 0; h < 2; ++h) {
+            const int r = 2 * rr + h;
+            float ds = 0.f;
+            if (k_base + 16 * kk + seg * 4 + r < T) {
+              const float pv =
+                  __expf(st_frag[kk][r] * scale - lse_q[qs]);
+              ds = scale * pv * (dpt_frag[kk][r] - delta_q[qs]);
+            }
+            packed |= (unsigned)f32_to_bf16(ds) << (16 * h);
           }
-          sm.ds_tile[wave][qs][seg * 4 + r][kk * 16 + col] =
-              (short)f32_to_bf16(ds);
+          pk[kk][rr] = packed;
         }
       }
-    }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-    // dQ += dS K
+      // redistribute dS to the MFMA A layout (same shuffle pattern as
+      // the forward's P)
+      const int kk_s = seg >> 1;
+      unsigned da_u[4];
 #pragma unroll
-    for (int qs = 0; qs < kQSub; ++qs) {
-      bf16x8 ds_frag = *reinterpret_cast<const bf16x8*>(
-          &sm.ds_tile[wave][qs][col][seg * 8]);
+      for (int j2 = 0; j2 < 4; ++j2) {
+        const int srcl = col + 16 * ((seg & 1) * 2 + (j2 >> 1));
+        const unsigned lo = (unsigned)__shfl((int)pk[0][j2 & 1], srcl);
+        const unsigned hi = (unsigned)__shfl((int)pk[1][j2 & 1], srcl);
+        da_u[j2] = kk_s ? hi : lo;
+      }
+      bf16x8 ds_frag;
+      {
+        union {
+          unsigned u[4];
+          bf16x8 v;
+        } cvt;
+        cvt.u[0] = da_u[0];
+        cvt.u[1] = da_u[1];
+        cvt.u[2] = da_u[2];
+        cvt.u[3] = da_u[3];
+        ds_frag = cvt.v;
+      }
+
+      // dQ += dS K
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
