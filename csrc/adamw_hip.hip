@@ -48,7 +48,6 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
   const float decay = 1.f - lr * wd;
   const float step_size = lr / bias_c1;
   const float inv_sqrt_c2 = rsqrtf(bias_c2);
-  const long stride = (long)gridDim.x * kBlock * 4;
   for (int ti = 0; ti < tab.count; ++ti) {
     float4* p = (float4*)tab.p[ti];
     const float4* g = (const float4*)tab.g[ti];
@@ -86,7 +85,6 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
       tab.v[ti][i] = vj;
     }
   }
-  (void)stride;
 }
 
 __global__ void mt_sqnorm_kernel(TensorTable tab, float* __restrict__ out) {

@@ -12,17 +12,17 @@
 // profiles/).
 //
 // Structure (forward):
-//   grid = (ceil(T/64), B*H); block = 256 threads = 4 waves.
-//   Wave w owns 16 query rows.  Per 32-key tile:
-//     - K tile [32][Dpad] and transposed V tile [D][32] staged in LDS
-//       (padded row strides, conflict-checked for ds_read_b128 groups),
-//     - S = Q K^T via __builtin_amdgcn_mfma_f32_16x16x32_bf16
-//       (2 k-subtiles x D/32 chunks), fp32 accumulators,
-//     - online softmax in registers (row state lives in the C-fragment
-//       lane groups; cross-lane row reduce = 4 x shfl_xor over the
-//       16-lane column group),
-//     - P routed through a small per-wave LDS tile to re-shape the
-//       softmax output (C layout) into the MFMA A layout for P V.
+//   grid = (B*H, ceil(T/QTile)); block = 256 threads = 4 waves; K/V
+//   tiles double-buffered in LDS with T14 split staging (next tile's
+//   HBM loads issue before this tile's MFMA phases), one barrier/tile.
+//   Per 32-key tile:
+//     - SWAPPED S^T = K Q^T via __builtin_amdgcn_mfma_f32_16x16x32_bf16
+//       (C-layout rows = k, cols = q: each lane owns ONE q row),
+//     - fully in-register online softmax (7 fmax + 2 cross-seg shuffles
+//       per row; state = two scalars per lane),
+//     - P re-enters the MFMA A layout via 4 packed bf16-pair shuffles
+//       (no LDS round-trip, no lgkmcnt drain),
+//     - O += P V against the transposed, bank-swizzled V tile.
 //   Epilogue: O /= rowsum, store bf16, write LSE (fp32) for backward.
 //
 // Backward: fully fused (FlashAttention-2 style), two MFMA kernels —
