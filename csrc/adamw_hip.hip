@@ -27,6 +27,7 @@ struct AdamWTable {
   const float* g[kMaxTensors];
   float* m[kMaxTensors];
   float* v[kMaxTensors];
+  unsigned short* mir[kMaxTensors];  // optional bf16 mirror of p
   long n[kMaxTensors];
   int count;
 };
@@ -71,6 +72,16 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
       p[i] = pv;
       m[i] = mv;
       v[i] = vv;
+      if (tab.mir[ti]) {
+        // keep the bf16 comm mirror in sync so the FSDP gather path
+        // never re-casts the fp32 master
+        struct Bf16x4 {
+          unsigned short d[4];
+        } mb;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) mb.d[j] = f32_to_bf16((&pv.x)[j]);
+        ((Bf16x4*)tab.mir[ti])[i] = mb;
+      }
     }
     // scalar tail (shards are padded to world_size, not necessarily to 4)
     const long tail = tab.n[ti] & 3;
@@ -83,6 +94,7 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
       tab.p[ti][i] = pj;
       tab.m[ti][i] = mj;
       tab.v[ti][i] = vj;
+      if (tab.mir[ti]) tab.mir[ti][i] = f32_to_bf16(pj);
     }
   }
 }
@@ -141,10 +153,14 @@ int grid_for(long total_elems) {
 void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> exp_avgs,
-                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                 std::vector<torch::Tensor> exp_avg_sqs,
+                 std::vector<c10::optional<torch::Tensor>> mirrors,
+                 double lr,
                  double beta1, double beta2, double eps, double weight_decay,
                  double bias_c1, double bias_c2,
                  c10::optional<torch::Tensor> grad_scale) {
+  TORCH_CHECK(mirrors.empty() || mirrors.size() == params.size(),
+              "mirrors must be empty or match params");
   const float* gs_ptr = nullptr;
   if (grad_scale.has_value()) {
     TORCH_CHECK(grad_scale->is_cuda() &&
@@ -167,6 +183,14 @@ void fused_adamw(std::vector<torch::Tensor> params,
       tab.g[i] = grads[base + i].data_ptr<float>();
       tab.m[i] = exp_avgs[base + i].data_ptr<float>();
       tab.v[i] = exp_avg_sqs[base + i].data_ptr<float>();
+      tab.mir[i] = nullptr;
+      if (!mirrors.empty() && mirrors[base + i].has_value()) {
+        auto& mt = *mirrors[base + i];
+        TORCH_CHECK(mt.scalar_type() == torch::kBFloat16 &&
+                        mt.numel() == p.numel() && mt.is_contiguous(),
+                    "mirror must be a contiguous bf16 tensor of p's size");
+        tab.mir[i] = (unsigned short*)mt.data_ptr();
+      }
       tab.n[i] = p.numel();
       total += tab.n[i];
     }

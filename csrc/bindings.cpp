@@ -24,7 +24,9 @@ std::vector<torch::Tensor> layernorm_add_bwd(torch::Tensor dy,
 void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> exp_avgs,
-                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                 std::vector<torch::Tensor> exp_avg_sqs,
+                 std::vector<c10::optional<torch::Tensor>> mirrors,
+                 double lr,
                  double beta1, double beta2, double eps, double weight_decay,
                  double bias_c1, double bias_c2,
                  c10::optional<torch::Tensor> grad_scale);

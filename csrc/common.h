@@ -16,6 +16,7 @@ using bf16_t = __hip_bfloat16;
 // 8 bf16 = 16 bytes, the coalescing sweet spot for bf16 streams
 typedef unsigned short ushort8_t __attribute__((ext_vector_type(8)));
 typedef float floatx4 __attribute__((ext_vector_type(4)));
+
 typedef short shortx8 __attribute__((ext_vector_type(8)));
 
 __device__ __forceinline__ float bf16_to_f32(unsigned short u) {

@@ -70,3 +70,50 @@ def test_fsdp_bf16_vs_fp32_one_step():
     diff = (out_bf - out_fp).abs().max().item()
     ref = out_fp.abs().max().item()
     assert diff < 0.05 * max(ref, 1.0), f"bf16/fp32 divergence {diff} vs {ref}"
+
+
+def test_bf16_mirror_invalidation_on_load():
+    """The AdamW-maintained bf16 comm mirror must be refreshed when new
+    master weights arrive via load_state_dict (otherwise gathers would
+    broadcast stale params)."""
+    from vit_10b_fsdp_example_amd.cli import parse_args
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
+    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    CommContext.reset()
+    args = [
+        "--fake_data", "--image_size", "28", "--patch_size", "14",
+        "--embed_dim", "64", "--num_heads", "2", "--num_blocks", "2",
+        "--num_classes", "10", "--batch_size", "4", "--num_workers", "0",
+    ]
+    device = xdist.init_distributed()
+    torch.manual_seed(0)
+    m1 = build_fsdp_vit_model(parse_args(args), device,
+                              compute_dtype=torch.bfloat16)
+    # train m1 a few steps so its weights differ from a fresh init
+    opt = FusedAdamW(m1.parameters(), lr=1e-2)
+    loss_fn = CrossEntropyLoss()
+    x = torch.randn(4, 3, 28, 28, device=device, dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (4,), device=device)
+    for _ in range(3):
+        loss_fn(m1(x), y).backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+    m1.eval()
+    with torch.no_grad():
+        out1 = m1(x)
+
+    torch.manual_seed(1)
+    m2 = build_fsdp_vit_model(parse_args(args), device,
+                              compute_dtype=torch.bfloat16)
+    m2.eval()
+    with torch.no_grad():
+        out_before = m2(x)
+    assert not torch.allclose(out_before.float(), out1.float(), atol=1e-3)
+    # loading m1's weights must invalidate m2's mirrors
+    m2.load_state_dict(m1.state_dict())
+    with torch.no_grad():
+        out2 = m2(x)
+    assert torch.allclose(out2.float(), out1.float(), rtol=1e-3, atol=1e-3)

@@ -71,11 +71,18 @@ class FusedAdamW(torch.optim.Optimizer):
                     p._deferred_grad_scale = None
 
             if params[0].is_cuda and ext() is not None:
+                # bf16 comm mirrors (FSDP attaches _bf16_mirror so the
+                # gather path never re-casts the fp32 master): the step
+                # kernel refreshes them in its epilogue
+                mirrors = [getattr(p, "_bf16_mirror", None) for p in params]
+                if not any(m is not None for m in mirrors):
+                    mirrors = []
                 ext().fused_adamw(
                     params,
                     grads,
                     exp_avgs,
                     exp_avg_sqs,
+                    mirrors,
                     group["lr"],
                     beta1,
                     beta2,
@@ -85,6 +92,9 @@ class FusedAdamW(torch.optim.Optimizer):
                     bias_c2,
                     grad_scale,
                 )
+                for p, m in zip(params, mirrors or []):
+                    if m is not None:
+                        p._mirror_fresh = True
             else:
                 if grad_scale is not None:
                     grads = torch._foreach_mul(grads, float(grad_scale))
@@ -93,6 +103,11 @@ class FusedAdamW(torch.optim.Optimizer):
                     group["lr"], beta1, beta2, group["eps"],
                     group["weight_decay"], bias_c1, bias_c2,
                 )
+                for p in params:
+                    m = getattr(p, "_bf16_mirror", None)
+                    if m is not None:
+                        m.copy_(p.detach().to(m.dtype))
+                        p._mirror_fresh = True
         return loss
 
     @staticmethod

@@ -147,6 +147,22 @@ class FullyShardedDataParallel(nn.Module):
         del flat
         self.flat_param = nn.Parameter(shard)
 
+        # bf16 comm mirror of the master shard: FusedAdamW refreshes it
+        # in its step epilogue (params carry _bf16_mirror/_mirror_fresh),
+        # so the gather path usually skips the fp32->bf16 cast entirely.
+        # Device-resident shards only; anything else falls back to the
+        # cast in _comm_shard.
+        self._mirror = None
+        if (compute_dtype == torch.bfloat16 and not shard_on_cpu
+                and self.device.type == "cuda"):
+            self._mirror = torch.empty(
+                self._shard_numel, dtype=torch.bfloat16, device=self.device
+            )
+            self._mirror.copy_(self.flat_param.data.to(torch.bfloat16))
+            self.flat_param._bf16_mirror = self._mirror
+            self.flat_param._mirror_fresh = True
+        self.register_load_state_dict_post_hook(self._invalidate_mirror_hook)
+
         # ---- full compute-dtype buffer + stable leaf views ----
         self._full_flat = torch.empty(
             self._padded_numel, dtype=compute_dtype, device=self.device
@@ -198,10 +214,22 @@ class FullyShardedDataParallel(nn.Module):
     # materialization
     # ------------------------------------------------------------------
 
+    @staticmethod
+    def _invalidate_mirror_hook(module, incompatible_keys):
+        # load_state_dict wrote new master values: the bf16 mirror is stale
+        if getattr(module, "_mirror", None) is not None:
+            module.flat_param._mirror_fresh = False
+
     def _comm_shard(self):
         """The shard in comm/compute dtype on the compute device (cast
         from the fp32 master; for --shard_on_cpu this is the pinned-host
-        -> device async staging copy)."""
+        -> device async staging copy).  Uses the AdamW-maintained bf16
+        mirror when it is fresh; refreshes it otherwise."""
+        if self._mirror is not None:
+            if not getattr(self.flat_param, "_mirror_fresh", False):
+                self._mirror.copy_(self.flat_param.data.to(torch.bfloat16))
+                self.flat_param._mirror_fresh = True
+            return self._mirror
         data = self.flat_param.data
         if data.device != self.device:
             data = data.to(self.device, non_blocking=True)
@@ -376,6 +404,10 @@ class FullyShardedDataParallel(nn.Module):
                 self.flat_param.grad = g
             else:
                 self.flat_param.grad.add_(g)
+            # the optimizer will rewrite the master next; unless it also
+            # refreshes the mirror (FusedAdamW does), the mirror is stale
+            if self._mirror is not None:
+                self.flat_param._mirror_fresh = False
         _free_storage(self._full_grad)
         self._grads_arrived = 0
 
