@@ -117,3 +117,33 @@ def test_bf16_mirror_invalidation_on_load():
     with torch.no_grad():
         out2 = m2(x)
     assert torch.allclose(out2.float(), out1.float(), rtol=1e-3, atol=1e-3)
+
+
+def test_long_sequence_448px_step():
+    """image_size 448 / patch 14 -> T=1024: the flash kernels' O(T)
+    memory path must handle the long-sequence configs end to end
+    (SURVEY §5 long-context: single-GPU large-T, no sequence sharding)."""
+    from vit_10b_fsdp_example_amd.cli import parse_args
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
+    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    CommContext.reset()
+    cfg = parse_args([
+        "--fake_data", "--image_size", "448", "--patch_size", "14",
+        "--embed_dim", "320", "--num_heads", "2", "--num_blocks", "2",
+        "--num_classes", "10", "--batch_size", "2", "--num_workers", "0",
+    ])
+    device = xdist.init_distributed()
+    torch.manual_seed(0)
+    model = build_fsdp_vit_model(cfg, device, compute_dtype=torch.bfloat16)
+    loss_fn = CrossEntropyLoss()
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    x = torch.randn(2, 3, 448, 448, device=device, dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (2,), device=device)
+    loss = loss_fn(model(x), y)
+    loss.backward()
+    model.clip_grad_norm_(1.0)
+    opt.step()
+    assert torch.isfinite(loss.float())
