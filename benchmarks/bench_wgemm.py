@@ -1,0 +1,36 @@
+#!/usr/bin/env python3
+"""wgrad GEMM (A^T B) vs hipBLASLt on the ViT-10B Linear-backward shapes."""
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+import vit_10b_fsdp_example_amd._C as C
+
+def bench(K, M, N, iters=15):
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    a = torch.randn(K, M, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(K, N, device=dev, dtype=torch.bfloat16)
+    (ours,) = C.wgrad_gemm(a, b, False)
+    ref = (a.float().t() @ b.float())
+    err = (ours.float() - ref).abs().max().item()
+    rel = err / ref.abs().max().item()
+
+    torch.cuda.synchronize(); t0 = time.time()
+    for _ in range(iters):
+        C.wgrad_gemm(a, b, False)
+    torch.cuda.synchronize(); t_ours = (time.time() - t0) / iters
+
+    torch.cuda.synchronize(); t0 = time.time()
+    for _ in range(iters):
+        torch.matmul(a.t(), b)
+    torch.cuda.synchronize(); t_lib = (time.time() - t0) / iters
+
+    fl = 2.0 * K * M * N
+    print(f"K{K} M{M} N{N}: ours {t_ours*1e3:7.2f} ms {fl/t_ours/1e12:7.0f} TF | "
+          f"hipBLASLt {t_lib*1e3:7.2f} ms {fl/t_lib/1e12:7.0f} TF | rel_err {rel:.4f}")
+
+if __name__ == "__main__":
+    bench(32768, 15360, 5120)   # qkv wgrad
+    bench(32768, 5120, 5120)    # proj wgrad
+    bench(32768, 20480, 5120)   # fc1 wgrad
+    bench(32768, 5120, 20480)   # fc2 wgrad

@@ -52,6 +52,9 @@ torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
                            torch::Tensor o, torch::Tensor lse,
                            long num_heads, double scale);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
+torch::Tensor tr16_probe(long mode);
+std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
+                                      bool with_bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
@@ -77,4 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fmha_bwd_qkv", &fmha_bwd_qkv,
         "flash attention backward producing the fused dqkv [B,T,3,H,D]");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
+  m.def("wgrad_gemm", &wgrad_gemm,
+        "C = A^T B weight-gradient GEMM (bf16, tr16 transpose reads)");
 }

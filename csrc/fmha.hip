@@ -859,6 +859,40 @@ __global__ __launch_bounds__(256) void fmha_dsoftmax_kernel(
   }
 }
 
+
+// ds_read_b64_tr_b16 semantics probe: fill LDS with the identity
+// pattern lds[i] = i (as raw u16), issue one transpose-read per lane at
+// a caller-chosen per-lane base expression, return each lane's 4
+// elements.  The GPU test decodes the (lane, elem) -> lds index map so
+// kernels can rely on it (no public ISA doc in this environment).
+__global__ __launch_bounds__(64) void tr16_probe_kernel(
+    unsigned short* __restrict__ out, int mode) {
+  HIP_DYNAMIC_SHARED(char, smem_raw)
+  unsigned short* lds = reinterpret_cast<unsigned short*>(smem_raw);
+  for (int i = threadIdx.x; i < 1024; i += 64) lds[i] = (unsigned short)i;
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  unsigned addr_elems;
+  switch (mode) {
+    case 0: addr_elems = 0; break;                                // uniform
+    case 1: addr_elems = (lane & 15) + (lane >> 4) * 64; break;   // guide map
+    case 2: addr_elems = (lane & 15) * 4; break;                  // 4/lane
+    default: addr_elems = lane * 4; break;
+  }
+  // byte address into LDS (dynamic region starts at 0: no static smem)
+  unsigned addr = addr_elems * 2 + (unsigned)__builtin_amdgcn_groupstaticsize();
+  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
+  u32x2 v;
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+               : "=v"(v)
+               : "v"(addr));
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  out[lane * 4 + 0] = (unsigned short)(v[0] & 0xffff);
+  out[lane * 4 + 1] = (unsigned short)(v[0] >> 16);
+  out[lane * 4 + 2] = (unsigned short)(v[1] & 0xffff);
+  out[lane * 4 + 3] = (unsigned short)(v[1] >> 16);
+}
+
 // Layout probe: one wave computes a single 16x16x32 MFMA from row-major
 // fp32 A[16][32], B[32][16] using the documented fragment maps; the GPU
 // test compares against torch.matmul to pin the layout assumptions.
@@ -1111,6 +1145,18 @@ torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
       D, launch_fmha_bwd<kD>(a, lse, delta, dbase, dbase + (long)H * D,
                              dbase + 2L * H * D));
   return dqkv;
+}
+
+
+torch::Tensor tr16_probe(long mode) {
+  auto out = torch::zeros({64, 4}, torch::TensorOptions()
+                                       .dtype(torch::kInt16)
+                                       .device(torch::kCUDA));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 2048, stream,
+                     (unsigned short*)out.data_ptr(), (int)mode);
+  HIP_CHECK_LAST();
+  return out;
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {

@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "csrc/adamw.hip",
         "csrc/cross_entropy.hip",
         "csrc/fmha.hip",
+        "csrc/wgemm.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
