@@ -29,7 +29,10 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import dist as xdist
-from ..ops import LayerNorm, attention_qkv, cross_entropy, fused_add_layer_norm  # noqa: F401
+from ..ops import (  # noqa: F401
+    LayerNorm, NativeLinear, attention_qkv, cross_entropy,
+    fused_add_layer_norm,
+)
 
 
 def init_vit_weights(module, recursive=False):
@@ -94,9 +97,11 @@ class Attention(nn.Module):
         self.num_heads = num_heads
         self.head_dim = dim // num_heads
         self.scale = self.head_dim ** -0.5
-        self.qkv = nn.Linear(dim, dim * 3, bias=qkv_bias)
+        # NativeLinear: weight gradients route through the csrc wgrad
+        # kernel where the shape qualifies (ops/linear.py)
+        self.qkv = NativeLinear(dim, dim * 3, bias=qkv_bias)
         self.attn_drop_p = attn_drop
-        self.proj = nn.Linear(dim, dim)
+        self.proj = NativeLinear(dim, dim)
         self.proj_drop = nn.Dropout(proj_drop)
 
     def forward(self, x):
@@ -120,8 +125,8 @@ class Mlp(nn.Module):
 
     def __init__(self, dim, hidden_dim, drop=0.0):
         super().__init__()
-        self.fc1 = nn.Linear(dim, hidden_dim)
-        self.fc2 = nn.Linear(hidden_dim, dim)
+        self.fc1 = NativeLinear(dim, hidden_dim)
+        self.fc2 = NativeLinear(hidden_dim, dim)
         self.drop = nn.Dropout(drop)
 
     def forward(self, x):

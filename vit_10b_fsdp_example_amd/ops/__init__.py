@@ -4,6 +4,7 @@ from .attention import attention, attention_qkv, math_attention
 from .cross_entropy import CrossEntropyLoss, cross_entropy
 from .adamw import FusedAdamW
 from .multi_tensor import local_sqnorm, scale_
+from .linear import NativeLinear
 
 __all__ = [
     "ext",
@@ -20,4 +21,5 @@ __all__ = [
     "FusedAdamW",
     "local_sqnorm",
     "scale_",
+    "NativeLinear",
 ]

@@ -1,0 +1,71 @@
+"""Linear layer with a native weight-gradient kernel.
+
+Forward and the input-gradient GEMM go through hipBLASLt (the sanctioned
+library path for plain GEMMs); the weight gradient dW = dY^T X — whose
+transposed-A layout hipBLASLt runs ~35-40% below its forward rate on the
+qkv/proj shapes — uses our csrc/wgemm.hip kernel built on gfx950's
+ds_read_b64_tr_b16 hardware transpose reads, with the bias gradient
+(column sums of dY) fused into the same kernel.
+
+Dispatch is shape-gated by measured crossover (benchmarks/bench_wgemm.py
+on MI355X):
+    shape (M=out, N=in, K=32768)    ours    hipBLASLt
+    qkv   15360 x 5120               931        660  TF/s
+    proj   5120 x 5120               884        825
+    fc1   20480 x 5120               898       1076   -> library
+    fc2    5120 x 20480              918       1080   -> library
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ._extension import ext, use_hip
+
+
+def _use_native_wgrad(dy2, x2, w):
+    if not (dy2.is_cuda and dy2.dtype == torch.bfloat16):
+        return False
+    m, n = w.shape[0], w.shape[1]
+    k = dy2.shape[0]
+    if not (k % 64 == 0 and m % 256 == 0 and n % 256 == 0):
+        return False
+    # measured crossover: hipBLASLt wins on the very wide/tall shapes
+    if m > 16384 or n > 8192:
+        return False
+    return use_hip(dy2) and hasattr(ext(), "wgrad_gemm")
+
+
+class _NativeLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return F.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = torch.matmul(dy, weight)  # hipBLASLt dgrad
+
+        if _use_native_wgrad(dy2, x2, weight):
+            outs = ext().wgrad_gemm(
+                dy2.contiguous(), x2.contiguous(), ctx.has_bias
+            )
+            dw = outs[0]
+            db = outs[1].to(dy.dtype) if ctx.has_bias else None
+        else:
+            dw = torch.matmul(dy2.t(), x2)
+            db = dy2.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+class NativeLinear(nn.Linear):
+    """Drop-in nn.Linear whose backward routes the weight gradient
+    through the native wgrad kernel when the shape qualifies (identical
+    state_dict keys and initialization)."""
+
+    def forward(self, x):
+        return _NativeLinearFn.apply(x, self.weight, self.bias)
