@@ -316,10 +316,13 @@ def test_wgrad_gemm_vs_reference():
                                rtol=0.05)
 
 
-def test_native_linear_grads():
+def test_native_linear_grads(monkeypatch):
     """NativeLinear's backward (native wgrad + fused dbias) matches
-    nn.Linear's autograd."""
+    nn.Linear's autograd (native path forced on for the test)."""
     from vit_10b_fsdp_example_amd.ops import NativeLinear
+    from vit_10b_fsdp_example_amd.ops import linear as linear_mod
+
+    monkeypatch.setattr(linear_mod, "_NATIVE_WGRAD", True)
 
     torch.manual_seed(6)
     K, N, M = 1024, 256, 512  # qualifies for the native path

@@ -1,20 +1,29 @@
 """Linear layer with a native weight-gradient kernel.
 
 Forward and the input-gradient GEMM go through hipBLASLt (the sanctioned
-library path for plain GEMMs); the weight gradient dW = dY^T X — whose
-transposed-A layout hipBLASLt runs ~35-40% below its forward rate on the
-qkv/proj shapes — uses our csrc/wgemm.hip kernel built on gfx950's
-ds_read_b64_tr_b16 hardware transpose reads, with the bias gradient
-(column sums of dY) fused into the same kernel.
+library path for plain GEMMs); the weight gradient dW = dY^T X can route
+through csrc/wgemm.hip (gfx950 ds_read_b64_tr_b16 hardware transpose
+reads, fused dbias column sums).
 
-Dispatch is shape-gated by measured crossover (benchmarks/bench_wgemm.py
-on MI355X):
-    shape (M=out, N=in, K=32768)    ours    hipBLASLt
+Measured ISOLATED (benchmarks/bench_wgemm.py, K=32768):
+    shape (M=out, N=in)             ours    hipBLASLt
     qkv   15360 x 5120               931        660  TF/s
     proj   5120 x 5120               884        825
-    fc1   20480 x 5120               898       1076   -> library
-    fc2    5120 x 20480              918       1080   -> library
+    fc1   20480 x 5120               898       1076
+    fc2    5120 x 20480              918       1080
+
+IN CONTEXT however (full training step, gpurun_out/prof_nl) the kernel
+measured ~520 TF/s — the isolated numbers were flattered by partial L3
+(256 MB Infinity Cache) residency of the repeated operands across bench
+iterations, which a real step does not get, and hipBLASLt's kernels are
+less L3-dependent.  The native path is therefore OPT-IN
+(VITFSDP_NATIVE_WGRAD=1) until its cold-cache performance beats the
+library (ROADMAP.md: deeper pipelining / 8-phase schedule).  This is a
+worked example of guide rule "bench any reference on the same data &
+cache state as your own kernel".
 """
+
+import os
 
 import torch
 import torch.nn as nn
@@ -23,7 +32,12 @@ import torch.nn.functional as F
 from ._extension import ext, use_hip
 
 
+_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "0") == "1"
+
+
 def _use_native_wgrad(dy2, x2, w):
+    if not _NATIVE_WGRAD:
+        return False
     if not (dy2.is_cuda and dy2.dtype == torch.bfloat16):
         return False
     m, n = w.shape[0], w.shape[1]
