@@ -76,10 +76,16 @@ class _NativeLinearFn(torch.autograd.Function):
         return dx, dw, db
 
 
+_PLAIN_LINEAR = os.environ.get("VITFSDP_PLAIN_LINEAR", "0") == "1"
+
+
 class NativeLinear(nn.Linear):
     """Drop-in nn.Linear whose backward routes the weight gradient
     through the native wgrad kernel when the shape qualifies (identical
-    state_dict keys and initialization)."""
+    state_dict keys and initialization).  VITFSDP_PLAIN_LINEAR=1 A/Bs
+    against stock autograd."""
 
     def forward(self, x):
+        if _PLAIN_LINEAR:
+            return F.linear(x, self.weight, self.bias)
         return _NativeLinearFn.apply(x, self.weight, self.bias)
