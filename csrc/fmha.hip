@@ -53,6 +53,7 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
 
 constexpr int kBlockThreads = 256;
 constexpr int kKTile = 32;  // k rows per inner iteration
@@ -73,20 +74,24 @@ struct FmhaShapes {
   static constexpr int KStride = DP + 8;
   static constexpr int VStride = kKTile + 8;  // 40
   static constexpr int PStride = kKTile + 8;  // 40
+  static constexpr int VRow = D + 16;  // row-major V: 8B-aligned tr
+                                       // lanes, conflict-free groups
   struct Shared {
     short k_tile[2][kKTile][KStride];  // double-buffered
-    short v_tile[2][D][VStride];       // transposed [d][k], dbuf
+    short v_tile[2][kKTile][VRow];     // ROW-major [k][d], dbuf: PV reads
+                                       // it via ds_read_b64_tr_b16
   };
+  // The backward kernels keep ONLY row-major tiles: the operands that
+  // need K-contiguous fragments (dQ's K, dKV's Q/dO) are read with
+  // ds_read_b64_tr_b16 hardware transpose reads, so the element-wise
+  // transposed copies (and their bank-conflict swizzles) are gone.
   struct SharedDQ {
-    short k_tile[kKTile][KStride];  // row-major (S^T operand)
+    short k_tile[kKTile][KStride];  // row-major (S^T + dQ operands)
     short v_tile[kKTile][KStride];  // row-major (dP^T operand)
-    short kt_tile[D][VStride];      // transposed K (dQ operand)
   };
   struct SharedDKV {
-    short q_tile[32][KStride];    // row-major (S^T operand)
-    short do_tile[32][KStride];   // row-major (dP^T operand)
-    short qt_tile[D][VStride];    // transposed Q (dK operand)
-    short dot_tile[D][VStride];   // transposed dO (dV operand)
+    short q_tile[32][KStride];    // row-major (S^T + dK operands)
+    short do_tile[32][KStride];   // row-major (dP^T + dV operands)
     short pt_tile[4][16][PStride];   // P^T re-layout per wave
     short dst_tile[4][16][PStride];  // dS^T re-layout per wave
   };
@@ -235,11 +240,8 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
 #pragma unroll
     for (int i = 0; i < kVVecs; ++i) {
       if (v_kr[i] < kKTile) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          sm.v_tile[buf][v_dc[i] + j][tr_swz(v_dc[i] + j, v_kr[i])] =
-              stage[i][j];
-        }
+        *reinterpret_cast<bf16x8*>(&sm.v_tile[buf][v_kr[i]][v_dc[i]]) =
+            stage[i];
       }
     }
   };
@@ -374,14 +376,52 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
       }
 
       // ---- O += P V : NC chunks of 16 output columns ----
+      // B fragments come from the ROW-major V tile via
+      // ds_read_b64_tr_b16 (2 transpose reads per 16-column chunk,
+      // 2-deep ring with counted waits): the element-wise transposed
+      // store and its bank-conflict swizzle are gone entirely.
+      const unsigned v_lds_base =
+          (unsigned)__builtin_amdgcn_groupstaticsize() +
+          (unsigned)(offsetof(typename S::Shared, v_tile)) +
+          (unsigned)cur * (unsigned)sizeof(sm.v_tile[0]);
+      auto v_tr = [&](int c, int h) -> u32x2 {
+        const unsigned addr =
+            v_lds_base +
+            2u * ((unsigned)((seg * 8 + 4 * h + (col >> 2)) * S::VRow +
+                             c * 16 + (col & 3) * 4));
+        u32x2 r;
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+                     : "=v"(r)
+                     : "v"(addr));
+        return r;
+      };
+      u32x2 vf[2][2];
+#pragma unroll
+      for (int h = 0; h < 2; ++h) vf[0][h] = v_tr(0, h);
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
-        // B fragment: B[seg*8+j][col] = V^T[c*16+col][seg*8+j]
-        bf16x8 v_frag = *reinterpret_cast<const bf16x8*>(
-            &sm.v_tile[cur][c * 16 + col]
-                      [tr_swz(c * 16 + col, seg * 8)]);
+        const int curc = c & 1;
+        if (c + 1 < NC) {
+#pragma unroll
+          for (int h = 0; h < 2; ++h) vf[curc ^ 1][h] = v_tr(c + 1, h);
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(vf[curc][0]), "+v"(vf[curc][1])
+                       : [cnt] "i"(2)
+                       : "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(vf[curc][0]), "+v"(vf[curc][1])
+                       : [cnt] "i"(0)
+                       : "memory");
+        }
+        union {
+          u32x2 uu[2];
+          bf16x8 v;
+        } vc;
+        vc.uu[0] = vf[curc][0];
+        vc.uu[1] = vf[curc][1];
         o_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            pa, v_frag, o_acc[qs][c], 0, 0, 0);
+            pa, vc.v, o_acc[qs][c], 0, 0, 0);
       }
     }
 
@@ -505,11 +545,6 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
       if (l_kr[i] >= kKTile) continue;
       *reinterpret_cast<bf16x8*>(&sm.k_tile[l_kr[i]][l_dc[i]]) = kst[i];
       *reinterpret_cast<bf16x8*>(&sm.v_tile[l_kr[i]][l_dc[i]]) = vst[i];
-      if (l_dc[i] < D) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          sm.kt_tile[l_dc[i] + j][tr_swz(l_dc[i] + j, l_kr[i])] = kst[i][j];
-      }
     }
   };
 
@@ -590,13 +625,49 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
         ds_frag = cvt.v;
       }
 
-      // dQ += dS K
+      // dQ += dS K: B fragments from the row-major K tile via hardware
+      // transpose reads (2-deep chunk ring, counted waits)
+      const unsigned k_lds_base =
+          (unsigned)__builtin_amdgcn_groupstaticsize() +
+          (unsigned)offsetof(typename S::SharedDQ, k_tile);
+      auto k_tr = [&](int c, int h) -> u32x2 {
+        const unsigned addr =
+            k_lds_base +
+            2u * ((unsigned)((seg * 8 + 4 * h + (col >> 2)) * S::KStride +
+                             c * 16 + (col & 3) * 4));
+        u32x2 r;
+        asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+                     : "=v"(r)
+                     : "v"(addr));
+        return r;
+      };
+      u32x2 kf[2][2];
+#pragma unroll
+      for (int h = 0; h < 2; ++h) kf[0][h] = k_tr(0, h);
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
-        bf16x8 ktb = *reinterpret_cast<const bf16x8*>(
-            &sm.kt_tile[c * 16 + col][tr_swz(c * 16 + col, seg * 8)]);
+        const int curc = c & 1;
+        if (c + 1 < NC) {
+#pragma unroll
+          for (int h = 0; h < 2; ++h) kf[curc ^ 1][h] = k_tr(c + 1, h);
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(kf[curc][0]), "+v"(kf[curc][1])
+                       : [cnt] "i"(2)
+                       : "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(kf[curc][0]), "+v"(kf[curc][1])
+                       : [cnt] "i"(0)
+                       : "memory");
+        }
+        union {
+          u32x2 uu[2];
+          bf16x8 v;
+        } kc2;
+        kc2.uu[0] = kf[curc][0];
+        kc2.uu[1] = kf[curc][1];
         dq_acc[qs][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            ds_frag, ktb, dq_acc[qs][c], 0, 0, 0);
+            ds_frag, kc2.v, dq_acc[qs][c], 0, 0, 0);
       }
     }
 
@@ -623,7 +694,7 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
 }
 
 template <int D>
-__global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
+__global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -704,14 +775,6 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
       if (s_qr[i] >= kQStream) continue;
       *reinterpret_cast<bf16x8*>(&sm.q_tile[s_qr[i]][s_dc[i]]) = q_st[i];
       *reinterpret_cast<bf16x8*>(&sm.do_tile[s_qr[i]][s_dc[i]]) = do_st[i];
-      if (s_dc[i] < D) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          sm.qt_tile[s_dc[i] + j][tr_swz(s_dc[i] + j, s_qr[i])] = q_st[i][j];
-          sm.dot_tile[s_dc[i] + j][tr_swz(s_dc[i] + j, s_qr[i])] =
-              do_st[i][j];
-        }
-      }
     }
   };
 
@@ -769,17 +832,62 @@ __global__ __launch_bounds__(kBlockThreads) void fmha_bwd_dkv_kernel(
         &sm.pt_tile[wave][col][seg * 8]);
     bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
         &sm.dst_tile[wave][col][seg * 8]);
+    // dV += P^T dO, dK += dS^T Q: both B fragments come from the
+    // row-major tiles via hardware transpose reads (2-deep chunk ring)
+    const unsigned base0 = (unsigned)__builtin_amdgcn_groupstaticsize();
+    const unsigned q_lds_base =
+        base0 + (unsigned)offsetof(typename S::SharedDKV, q_tile);
+    const unsigned do_lds_base =
+        base0 + (unsigned)offsetof(typename S::SharedDKV, do_tile);
+    auto qdo_tr = [&](unsigned base, int c, int h) -> u32x2 {
+      const unsigned addr =
+          base + 2u * ((unsigned)((seg * 8 + 4 * h + (col >> 2)) * S::KStride +
+                                  c * 16 + (col & 3) * 4));
+      u32x2 r;
+      asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
+                   : "=v"(r)
+                   : "v"(addr));
+      return r;
+    };
+    u32x2 qf[2][2], dof[2][2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      dof[0][h] = qdo_tr(do_lds_base, 0, h);
+      qf[0][h] = qdo_tr(q_lds_base, 0, h);
+    }
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
-      const int swz = tr_swz(c * 16 + col, seg * 8);
-      bf16x8 dotb = *reinterpret_cast<const bf16x8*>(
-          &sm.dot_tile[c * 16 + col][swz]);
-      bf16x8 qtb = *reinterpret_cast<const bf16x8*>(
-          &sm.qt_tile[c * 16 + col][swz]);
+      const int curc = c & 1;
+      if (c + 1 < NC) {
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          dof[curc ^ 1][h] = qdo_tr(do_lds_base, c + 1, h);
+          qf[curc ^ 1][h] = qdo_tr(q_lds_base, c + 1, h);
+        }
+        asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                     : "+v"(dof[curc][0]), "+v"(dof[curc][1]),
+                       "+v"(qf[curc][0]), "+v"(qf[curc][1])
+                     : [cnt] "i"(4)
+                     : "memory");
+      } else {
+        asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                     : "+v"(dof[curc][0]), "+v"(dof[curc][1]),
+                       "+v"(qf[curc][0]), "+v"(qf[curc][1])
+                     : [cnt] "i"(0)
+                     : "memory");
+      }
+      union {
+        u32x2 uu[2];
+        bf16x8 v;
+      } dc2, qc2;
+      dc2.uu[0] = dof[curc][0];
+      dc2.uu[1] = dof[curc][1];
+      qc2.uu[0] = qf[curc][0];
+      qc2.uu[1] = qf[curc][1];
       dv_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          pt_frag, dotb, dv_acc[c], 0, 0, 0);
+          pt_frag, dc2.v, dv_acc[c], 0, 0, 0);
       dk_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          dst_frag, qtb, dk_acc[c], 0, 0, 0);
+          dst_frag, qc2.v, dk_acc[c], 0, 0, 0);
     }
 
     if (has_next) {
@@ -881,7 +989,6 @@ __global__ __launch_bounds__(64) void tr16_probe_kernel(
   }
   // byte address into LDS (dynamic region starts at 0: no static smem)
   unsigned addr = addr_elems * 2 + (unsigned)__builtin_amdgcn_groupstaticsize();
-  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
   u32x2 v;
   asm volatile("ds_read_b64_tr_b16 %0, %1 offset:0"
                : "=v"(v)
