@@ -72,7 +72,6 @@ struct FmhaShapes {
   // LDS row strides in elements, padded so the 16-lane ds_read_b128
   // column groups land on distinct banks (stride*2B stays 16B-aligned)
   static constexpr int KStride = DP + 8;
-  static constexpr int VStride = kKTile + 8;  // 40
   static constexpr int PStride = kKTile + 8;  // 40
   static constexpr int VRow = D + 16;  // row-major V: 8B-aligned tr
                                        // lanes, conflict-free groups
@@ -97,16 +96,6 @@ struct FmhaShapes {
   };
 };
 
-
-// Transposed LDS tiles ([d][k], k-stride 32+8) are written element-wise
-// (scalar b16 stores) during the tile transpose.  Without a swizzle,
-// lanes writing the same k for d-rows 8 apart land on ONE bank (row
-// stride 80 B == 0 mod 128 B): a ~20-way conflict.  XOR the k index
-// with bits >=3 of (d>>3): banks spread 4x, and the 16-byte-aligned
-// 8-element runs the b128 reads need stay contiguous.
-__device__ __forceinline__ int tr_swz(int d, int k) {
-  return k ^ (((d >> 3) & 3) << 3);
-}
 
 // grid = (B*H, ceil(T/kQTile)): bh on x so a head's q-tiles share the
 // XCD whose L2 already holds its K/V (blockIdx linearization round-
