@@ -56,7 +56,12 @@ def init_distributed(device_index=None, timeout_minutes=30):
         device = torch.device("cpu")
         backend = "gloo"
 
-    if world_size > 1 and not dist.is_initialized():
+    # initialize the process group whenever a launcher provided topology
+    # env (torchrun sets WORLD_SIZE even for one rank): this keeps the
+    # RCCL/gloo communicator-creation path identical between N=1 smoke
+    # runs and the real multi-GPU launches
+    launched = "WORLD_SIZE" in os.environ
+    if (world_size > 1 or launched) and not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group(
