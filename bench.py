@@ -167,6 +167,9 @@ def main():
         }
         print(json.dumps(result), flush=True)
 
+    if xdist.is_distributed():
+        torch.distributed.destroy_process_group()
+
 
 if __name__ == "__main__":
     main()

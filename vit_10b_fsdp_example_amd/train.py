@@ -209,3 +209,7 @@ def main(cfg):
     xdist.master_print(f"device: {device}, world size: {xdist.get_world_size()}")
     train(cfg)
     xdist.master_print("training completed")
+    if xdist.is_distributed():
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
