@@ -167,10 +167,15 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
+      if (valid && d0 + 8 <= D) {
+        // contiguous 8 elements: one 16-B load instead of 8 scalars
+        q_frag[qs][kc] = *reinterpret_cast<const bf16x8*>(&q[base + d0]);
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        q_frag[qs][kc][j] =
-            (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
+        for (int j = 0; j < 8; ++j) {
+          q_frag[qs][kc][j] =
+              (valid && d0 + j < D) ? q[base + d0 + j] : (short)0;
+        }
       }
     }
   }
@@ -485,11 +490,17 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
+      if (valid && d0 + 8 <= D) {
+        q_frag[qs][kc] = *reinterpret_cast<const bf16x8*>(&q[base + d0]);
+        do_frag[qs][kc] =
+            *reinterpret_cast<const bf16x8*>(&dout[dbase + d0]);
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const bool ok = valid && d0 + j < D;
-        q_frag[qs][kc][j] = ok ? q[base + d0 + j] : (short)0;
-        do_frag[qs][kc][j] = ok ? dout[dbase + d0 + j] : (short)0;
+        for (int j = 0; j < 8; ++j) {
+          const bool ok = valid && d0 + j < D;
+          q_frag[qs][kc][j] = ok ? q[base + d0 + j] : (short)0;
+          do_frag[qs][kc][j] = ok ? dout[dbase + d0 + j] : (short)0;
+        }
       }
     }
     // per-lane row stats: this lane's q row is q = col (swapped layout)
@@ -717,11 +728,16 @@ __global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       const int d0 = kc * 32 + seg * 8;
+      if (valid && d0 + 8 <= D) {
+        k_frag[kc] = *reinterpret_cast<const bf16x8*>(&k[base + d0]);
+        v_frag[kc] = *reinterpret_cast<const bf16x8*>(&v[base + d0]);
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const bool ok = valid && d0 + j < D;
-        k_frag[kc][j] = ok ? k[base + d0 + j] : (short)0;
-        v_frag[kc][j] = ok ? v[base + d0 + j] : (short)0;
+        for (int j = 0; j < 8; ++j) {
+          const bool ok = valid && d0 + j < D;
+          k_frag[kc][j] = ok ? k[base + d0 + j] : (short)0;
+          v_frag[kc][j] = ok ? v[base + d0 + j] : (short)0;
+        }
       }
     }
   }
@@ -822,8 +838,11 @@ __global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
         &sm.pt_tile[wave][col][seg * 8]);
     bf16x8 dst_frag = *reinterpret_cast<const bf16x8*>(
         &sm.dst_tile[wave][col][seg * 8]);
-    // dV += P^T dO, dK += dS^T Q: both B fragments come from the
-    // row-major tiles via hardware transpose reads (2-deep chunk ring)
+    // dV += P^T dO, dK += dS^T Q: B fragments from the row-major tiles
+    // via hardware transpose reads.  TWO sequential passes (dV then dK),
+    // each with a single-operand 2-deep ring: the combined dual ring
+    // costed 8 more live registers and pushed the kernel to 266 regs /
+    // 1 wave/SIMD.
     const unsigned base0 = (unsigned)__builtin_amdgcn_groupstaticsize();
     const unsigned q_lds_base =
         base0 + (unsigned)offsetof(typename S::SharedDKV, q_tile);
@@ -839,47 +858,41 @@ __global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
                    : "v"(addr));
       return r;
     };
-    u32x2 qf[2][2], dof[2][2];
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      dof[0][h] = qdo_tr(do_lds_base, 0, h);
-      qf[0][h] = qdo_tr(q_lds_base, 0, h);
-    }
+    for (int pass = 0; pass < 2; ++pass) {
+      const unsigned tr_base = pass == 0 ? do_lds_base : q_lds_base;
+      const bf16x8 a_frag = pass == 0 ? pt_frag : dst_frag;
+      f32x4* accs = pass == 0 ? dv_acc : dk_acc;
+      u32x2 bfrag[2][2];
 #pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      const int curc = c & 1;
-      if (c + 1 < NC) {
+      for (int h = 0; h < 2; ++h) bfrag[0][h] = qdo_tr(tr_base, 0, h);
 #pragma unroll
-        for (int h = 0; h < 2; ++h) {
-          dof[curc ^ 1][h] = qdo_tr(do_lds_base, c + 1, h);
-          qf[curc ^ 1][h] = qdo_tr(q_lds_base, c + 1, h);
+      for (int c = 0; c < NC; ++c) {
+        const int curc = c & 1;
+        if (c + 1 < NC) {
+#pragma unroll
+          for (int h = 0; h < 2; ++h)
+            bfrag[curc ^ 1][h] = qdo_tr(tr_base, c + 1, h);
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(bfrag[curc][0]), "+v"(bfrag[curc][1])
+                       : [cnt] "i"(2)
+                       : "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(bfrag[curc][0]), "+v"(bfrag[curc][1])
+                       : [cnt] "i"(0)
+                       : "memory");
         }
-        asm volatile("s_waitcnt lgkmcnt(%[cnt])"
-                     : "+v"(dof[curc][0]), "+v"(dof[curc][1]),
-                       "+v"(qf[curc][0]), "+v"(qf[curc][1])
-                     : [cnt] "i"(4)
-                     : "memory");
-      } else {
-        asm volatile("s_waitcnt lgkmcnt(%[cnt])"
-                     : "+v"(dof[curc][0]), "+v"(dof[curc][1]),
-                       "+v"(qf[curc][0]), "+v"(qf[curc][1])
-                     : [cnt] "i"(0)
-                     : "memory");
+        union {
+          u32x2 uu[2];
+          bf16x8 v;
+        } bc2;
+        bc2.uu[0] = bfrag[curc][0];
+        bc2.uu[1] = bfrag[curc][1];
+        accs[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, bc2.v,
+                                                          accs[c], 0, 0, 0);
       }
-      union {
-        u32x2 uu[2];
-        bf16x8 v;
-      } dc2, qc2;
-      dc2.uu[0] = dof[curc][0];
-      dc2.uu[1] = dof[curc][1];
-      qc2.uu[0] = qf[curc][0];
-      qc2.uu[1] = qf[curc][1];
-      dv_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          pt_frag, dc2.v, dv_acc[c], 0, 0, 0);
-      dk_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          dst_frag, qc2.v, dk_acc[c], 0, 0, 0);
     }
-
     if (has_next) {
       __syncthreads();  // everyone is done reading tile qt
       write_qdo_tiles();
