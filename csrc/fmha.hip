@@ -21,7 +21,8 @@
 //       per row; state = two scalars per lane),
 //     - P re-enters the MFMA A layout via 4 packed bf16-pair shuffles
 //       (no LDS round-trip, no lgkmcnt drain),
-//     - O += P V against the transposed, bank-swizzled V tile.
+//     - O += P V with B fragments gathered from the ROW-major V tile
+//       by ds_read_b64_tr_b16 hardware transpose reads (2-deep ring).
 //   Epilogue: O /= rowsum, store bf16, write LSE (fp32) for backward.
 //
 // Backward: fully fused (FlashAttention-2 style), two MFMA kernels —
