@@ -5,24 +5,33 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 import vit_10b_fsdp_example_amd._C as C
 
-def bench(K, M, N, iters=15):
+def bench(K, M, N, iters=12):
+    """COLD-CACHE methodology: rotate 4 independent operand sets so the
+    256 MB Infinity Cache cannot keep the operands warm between
+    iterations (a single reused set flattered the custom kernel ~1.8x
+    vs its in-training-step performance; see ops/linear.py)."""
     dev = torch.device("cuda", 0)
     torch.manual_seed(0)
-    a = torch.randn(K, M, device=dev, dtype=torch.bfloat16)
-    b = torch.randn(K, N, device=dev, dtype=torch.bfloat16)
+    sets = [
+        (torch.randn(K, M, device=dev, dtype=torch.bfloat16),
+         torch.randn(K, N, device=dev, dtype=torch.bfloat16))
+        for _ in range(4)
+    ]
+    a, b = sets[0]
     (ours,) = C.wgrad_gemm(a, b, False)
     ref = (a.float().t() @ b.float())
-    err = (ours.float() - ref).abs().max().item()
-    rel = err / ref.abs().max().item()
+    rel = (ours.float() - ref).abs().max().item() / ref.abs().max().item()
 
     torch.cuda.synchronize(); t0 = time.time()
-    for _ in range(iters):
-        C.wgrad_gemm(a, b, False)
+    for i in range(iters):
+        aa, bb = sets[i % 4]
+        C.wgrad_gemm(aa, bb, False)
     torch.cuda.synchronize(); t_ours = (time.time() - t0) / iters
 
     torch.cuda.synchronize(); t0 = time.time()
-    for _ in range(iters):
-        torch.matmul(a.t(), b)
+    for i in range(iters):
+        aa, bb = sets[i % 4]
+        torch.matmul(aa.t(), bb)
     torch.cuda.synchronize(); t_lib = (time.time() - t0) / iters
 
     fl = 2.0 * K * M * N
