@@ -31,11 +31,18 @@ def bench(K, M, N, iters=12):
     torch.cuda.synchronize(); t0 = time.time()
     for i in range(iters):
         aa, bb = sets[i % 4]
+        C.wgrad_gemm(aa, bb, True)
+    torch.cuda.synchronize(); t_bias = (time.time() - t0) / iters
+
+    torch.cuda.synchronize(); t0 = time.time()
+    for i in range(iters):
+        aa, bb = sets[i % 4]
         torch.matmul(aa.t(), bb)
     torch.cuda.synchronize(); t_lib = (time.time() - t0) / iters
 
     fl = 2.0 * K * M * N
     print(f"K{K} M{M} N{N}: ours {t_ours*1e3:7.2f} ms {fl/t_ours/1e12:7.0f} TF | "
+          f"ours+bias {t_bias*1e3:7.2f} ms {fl/t_bias/1e12:7.0f} TF | "
           f"hipBLASLt {t_lib*1e3:7.2f} ms {fl/t_lib/1e12:7.0f} TF | rel_err {rel:.4f}")
 
 if __name__ == "__main__":

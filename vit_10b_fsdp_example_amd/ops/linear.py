@@ -65,11 +65,11 @@ class _NativeLinearFn(torch.autograd.Function):
         dx = torch.matmul(dy, weight)  # hipBLASLt dgrad
 
         if _use_native_wgrad(dy2, x2, weight):
-            outs = ext().wgrad_gemm(
-                dy2.contiguous(), x2.contiguous(), ctx.has_bias
-            )
-            dw = outs[0]
-            db = outs[1].to(dy.dtype) if ctx.has_bias else None
+            # dbias via the fused in-kernel column sums measured ~1.7x
+            # slower end-to-end (per-fragment unpack+accumulate VALU);
+            # the plain reduction is one cheap memory-bound pass
+            (dw,) = ext().wgrad_gemm(dy2.contiguous(), x2.contiguous(), False)
+            db = dy2.sum(0) if ctx.has_bias else None
         else:
             dw = torch.matmul(dy2.t(), x2)
             db = dy2.sum(0) if ctx.has_bias else None
