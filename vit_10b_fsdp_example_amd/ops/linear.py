@@ -12,15 +12,14 @@ Measured ISOLATED (benchmarks/bench_wgemm.py, K=32768):
     fc1   20480 x 5120               898       1076
     fc2    5120 x 20480              918       1080
 
-IN CONTEXT however (full training step, gpurun_out/prof_nl) the kernel
-measured ~520 TF/s — the isolated numbers were flattered by partial L3
-(256 MB Infinity Cache) residency of the repeated operands across bench
-iterations, which a real step does not get, and hipBLASLt's kernels are
-less L3-dependent.  The native path is therefore OPT-IN
-(VITFSDP_NATIVE_WGRAD=1) until its cold-cache performance beats the
-library (ROADMAP.md: deeper pipelining / 8-phase schedule).  This is a
-worked example of guide rule "bench any reference on the same data &
-cache state as your own kernel".
+The first in-step measurement showed ~520 TF/s and was initially
+misattributed to warm-cache flattery; a cold-cache bench (4 rotating
+operand sets past the 256 MB L3) plus a with/without-bias A/B pinned
+the real cause: the FUSED dbias column sums (per-fragment unpack +
+accumulate VALU) cost the kernel ~1.7x (945 -> 535 TF on the qkv
+shape).  dbias is now a separate memory-bound reduction and the native
+path is ON by default for the shapes where it wins
+(VITFSDP_NATIVE_WGRAD=0 disables).
 """
 
 import os
@@ -32,7 +31,7 @@ import torch.nn.functional as F
 from ._extension import ext, use_hip
 
 
-_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "0") == "1"
+_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "1") == "1"
 
 
 def _use_native_wgrad(dy2, x2, w):
@@ -81,10 +80,12 @@ class NativeLinear(nn.Linear):
     native wgrad kernel when VITFSDP_NATIVE_WGRAD=1 and the shape
     qualifies (identical state_dict keys and initialization).
 
-    Default is stock autograd: a within-box A/B showed the custom
-    Function with the library wgrad formulation costs ~126 ms/step at
-    ViT-10B (56.7 -> 53.7 img/s) — torch's own matmul backward picks a
-    faster GEMM layout for dW than the explicit dy^T @ x call."""
+    When the native path is disabled it falls back to stock autograd
+    entirely: a within-box A/B showed the custom Function with the
+    LIBRARY wgrad formulation costs ~126 ms/step at ViT-10B — torch's
+    own matmul backward picks a faster dW GEMM layout than the explicit
+    dy^T @ x call — so there is no reason to take the Function without
+    the native kernel."""
 
     def forward(self, x):
         if _NATIVE_WGRAD:
