@@ -12,14 +12,19 @@ Measured ISOLATED (benchmarks/bench_wgemm.py, K=32768):
     fc1   20480 x 5120               898       1076
     fc2    5120 x 20480              918       1080
 
-The first in-step measurement showed ~520 TF/s and was initially
-misattributed to warm-cache flattery; a cold-cache bench (4 rotating
-operand sets past the 256 MB L3) plus a with/without-bias A/B pinned
-the real cause: the FUSED dbias column sums (per-fragment unpack +
-accumulate VALU) cost the kernel ~1.7x (945 -> 535 TF on the qkv
-shape).  dbias is now a separate memory-bound reduction and the native
-path is ON by default for the shapes where it wins
-(VITFSDP_NATIVE_WGRAD=0 disables).
+Measurement history (all on MI355X, receipts in profiles/PROFILES.md):
+  * first in-step number ~520 TF/s was caused by the FUSED dbias column
+    sums (per-fragment unpack+accumulate VALU, 945 -> 535 TF on the qkv
+    shape); dbias is now a separate memory-bound reduction;
+  * a cold-cache bench (4 rotating operand sets past the 256 MB L3)
+    confirmed the GEMM core itself beats hipBLASLt isolated;
+  * but a within-box end-to-end A/B still reads 54.3 vs 58.0 img/s with
+    the custom autograd Function active: replacing torch's fused linear
+    backward costs more on the OTHER gradients (dX formulation /
+    kernel-selection) than the native dW saves.
+The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1); closing
+the Function-overhead gap (e.g. an addmm-epilogue dX or grabbing
+torch's exact backward kernels) is ROADMAP work.
 """
 
 import os
@@ -31,7 +36,7 @@ import torch.nn.functional as F
 from ._extension import ext, use_hip
 
 
-_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "1") == "1"
+_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "0") == "1"
 
 
 def _use_native_wgrad(dy2, x2, w):
