@@ -19,12 +19,18 @@ Measurement history (all on MI355X, receipts in profiles/PROFILES.md):
   * a cold-cache bench (4 rotating operand sets past the 256 MB L3)
     confirmed the GEMM core itself beats hipBLASLt isolated;
   * but a within-box end-to-end A/B still reads 54.3 vs 58.0 img/s with
-    the custom autograd Function active: replacing torch's fused linear
-    backward costs more on the OTHER gradients (dX formulation /
-    kernel-selection) than the native dW saves.
-The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1); closing
-the Function-overhead gap (e.g. an addmm-epilogue dX or grabbing
-torch's exact backward kernels) is ROADMAP work.
+    the custom autograd Function active;
+  * per-layer isolation (benchmarks/bench_linear_bwd.py) shows stock,
+    Function-with-library-wgrad, and Function-with-native-wgrad all
+    EQUAL (qkv 12.9/12.6/12.9 ms fwd+bwd) — i.e. (a) torch's fused
+    linear backward already picks a dW kernel as fast as ours (the
+    standalone torch.matmul(a.t(), b) used as the bench baseline is the
+    SLOW formulation, not what training actually runs), and (b) the
+    end-to-end Function cost is an interaction with the grad-checkpoint
+    / FSDP context, not a per-layer kernel effect.
+The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1): it is a
+validated, shape-gated alternative at parity with the library, and the
+remaining work is the checkpoint-context interaction (ROADMAP).
 """
 
 import os
