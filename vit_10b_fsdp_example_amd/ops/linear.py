@@ -27,10 +27,22 @@ Measurement history (all on MI355X, receipts in profiles/PROFILES.md):
     standalone torch.matmul(a.t(), b) used as the bench baseline is the
     SLOW formulation, not what training actually runs), and (b) the
     end-to-end Function cost is an interaction with the grad-checkpoint
-    / FSDP context, not a per-layer kernel effect.
-The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1): it is a
-validated, shape-gated alternative at parity with the library, and the
-remaining work is the checkpoint-context interaction (ROADMAP).
+    context, not a per-layer kernel effect;
+  * ROOT-CAUSED via benchmarks/ktrace_diff.py on the ON/OFF traces
+    (profiles/PROFILES.md "ktrace diff") + a CPU dispatch-count repro
+    (tests/test_checkpoint_earlystop.py): torch's codegen'd addmm packs
+    its input SavedVariables BEFORE dispatching the kernel, so
+    checkpoint(use_reentrant=False) early-stops the recompute before
+    the region's LAST GEMM (fc2 forward is never recomputed).  A custom
+    autograd.Function packs only after forward returns, so wrapping the
+    linears in one forces that GEMM back into every block recompute:
+    +128 forward GEMMs (+657 ms) per 4 profiled steps, ~164 ms/step —
+    which is the previously "unexplained" part of the A/B gap.  The
+    rest of the gap was the since-removed bias-fused wgemm variant.
+The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1): the GEMM
+core is at parity, but any always-on integration must keep the stock
+autograd node for the forward (wgrad hooked below autograd, or the
+checkpoint region restructured) instead of substituting a Function.
 """
 
 import os
