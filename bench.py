@@ -46,7 +46,9 @@ def main():
     from vit_10b_fsdp_example_amd import dist as xdist
     from vit_10b_fsdp_example_amd.cli import parse_args as cli_parse
     from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
-    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.ops import (
+        CrossEntropyLoss, FusedAdamW, wgrad_backward_context,
+    )
     from vit_10b_fsdp_example_amd.utils import get_warmup_cosine_scheduler
 
     from vit_10b_fsdp_example_amd.tuning import enable_tunableop
@@ -99,7 +101,8 @@ def main():
     def one_step(i):
         x, y = batches[i % 2]
         loss = loss_fn(model(x), y)
-        loss.backward()
+        with wgrad_backward_context():
+            loss.backward()
         if cfg.clip_grad_norm > 0:
             model.clip_grad_norm_(cfg.clip_grad_norm, defer_scale=True)
         opt.step()

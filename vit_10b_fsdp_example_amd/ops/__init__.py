@@ -4,7 +4,7 @@ from .attention import attention, attention_qkv, math_attention
 from .cross_entropy import CrossEntropyLoss, cross_entropy
 from .adamw import FusedAdamW
 from .multi_tensor import local_sqnorm, scale_
-from .linear import NativeLinear
+from .linear import NativeLinear, NativeWgradMode, wgrad_backward_context
 
 __all__ = [
     "ext",
@@ -22,4 +22,6 @@ __all__ = [
     "local_sqnorm",
     "scale_",
     "NativeLinear",
+    "NativeWgradMode",
+    "wgrad_backward_context",
 ]

@@ -45,16 +45,24 @@ autograd node for the forward (wgrad hooked below autograd, or the
 checkpoint region restructured) instead of substituting a Function.
 """
 
+import contextlib
 import os
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+from torch.utils._python_dispatch import TorchDispatchMode
 
 from ._extension import ext, use_hip
 
 
-_NATIVE_WGRAD = os.environ.get("VITFSDP_NATIVE_WGRAD", "0") == "1"
+# "1": custom autograd.Function (kept for A/B history; defeats checkpoint
+#      early-stop, see tests/test_checkpoint_earlystop.py)
+# "2": dispatcher interception below autograd (NativeWgradMode) — the
+#      stock addmm node stays, early-stop is preserved
+_WGRAD_MODE = os.environ.get("VITFSDP_NATIVE_WGRAD", "0")
+_NATIVE_WGRAD = _WGRAD_MODE == "1"
+_NATIVE_WGRAD_DISPATCH = _WGRAD_MODE == "2"
 
 
 def _use_native_wgrad(dy2, x2, w):
@@ -96,6 +104,89 @@ class _NativeLinearFn(torch.autograd.Function):
             dw = torch.matmul(dy2.t(), x2)
             db = dy2.sum(0) if ctx.has_bias else None
         return dx, dw, db
+
+
+def _wgrad_mm_shapes_ok(m, n, k):
+    """Shape gate for the native wgrad kernel: the 256x256x64 tile must
+    divide evenly, and the measured ViT-10B crossover keeps the very
+    wide/tall MLP shapes (fc1/fc2, 20480-wide) on hipBLASLt while qkv
+    and proj go native.  Symmetric bound because the dispatcher sees dW
+    in whichever orientation AddmmBackward chose."""
+    return (k % 64 == 0 and m % 256 == 0 and n % 256 == 0
+            and m <= 16384 and n <= 16384)
+
+
+def _is_wgrad_mm(at, b):
+    """Match the weight-gradient GEMM AddmmBackward emits:
+    mm(saved_input.t(), grad_out) — first operand a transposed view of a
+    contiguous [K, M] base, second operand contiguous [K, N]."""
+    if at.dim() != 2 or b.dim() != 2 or at.shape[1] != b.shape[0]:
+        return False
+    m, k = at.shape
+    if m < 2 or k < 2:
+        return False  # degenerate strides are ambiguous
+    if not (at.stride(0) == 1 and at.stride(1) == m):
+        return False
+    if not (b.stride(1) == 1 and b.stride(0) == b.shape[1]):
+        return False
+    return _wgrad_mm_shapes_ok(m, b.shape[1], k)
+
+
+class NativeWgradMode(TorchDispatchMode):
+    """Reroute ONLY the dW GEMMs to csrc/wgemm.hip, below autograd.
+
+    This is the corrected integration after the Function-path root cause
+    (see module docstring): pushing a mode around ``loss.backward()``
+    leaves every forward an ordinary codegen'd addmm — non-reentrant
+    checkpoint early-stop keeps skipping the last recompute GEMM — and
+    the mode swaps the kernel at dispatch time when AddmmBackward runs
+    its mm(x^T, dy).  The engine propagates the mode into its worker
+    threads via ThreadLocalState, which
+    tests/test_native_wgrad_dispatch.py verifies on CPU.
+
+    ``handler(a, b)`` (a: contiguous [K, M] base, b: contiguous [K, N],
+    returns [M, N]) is a test seam; the default requires the HIP
+    extension and bf16 CUDA tensors.
+    """
+
+    def __init__(self, handler=None):
+        super().__init__()
+        self.hits = 0
+        self._handler = handler
+
+    def _route(self, at, b):
+        if self._handler is not None:
+            return self._handler(at.t(), b)
+        (dw,) = ext().wgrad_gemm(at.t().contiguous(), b.contiguous(), False)
+        return dw
+
+    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
+        kwargs = kwargs or {}
+        if (
+            func is torch.ops.aten.mm.default
+            and _is_wgrad_mm(*args)
+            and (
+                self._handler is not None
+                or (
+                    args[0].is_cuda
+                    and args[0].dtype == torch.bfloat16
+                    and args[1].dtype == torch.bfloat16
+                    and use_hip(args[0])
+                    and hasattr(ext(), "wgrad_gemm")
+                )
+            )
+        ):
+            self.hits += 1
+            return self._route(*args)
+        return func(*args, **kwargs)
+
+
+def wgrad_backward_context():
+    """Context manager for ``loss.backward()``: NativeWgradMode when
+    VITFSDP_NATIVE_WGRAD=2, otherwise a no-op."""
+    if _NATIVE_WGRAD_DISPATCH:
+        return NativeWgradMode()
+    return contextlib.nullcontext()
 
 
 class NativeLinear(nn.Linear):

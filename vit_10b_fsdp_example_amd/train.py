@@ -23,7 +23,7 @@ import torch
 from . import dist as xdist
 from .data import build_datasets
 from .models import build_fsdp_vit_model
-from .ops import CrossEntropyLoss, FusedAdamW
+from .ops import CrossEntropyLoss, FusedAdamW, wgrad_backward_context
 from .utils import SmoothedValue, get_warmup_cosine_scheduler, save_ckpt, load_ckpt
 
 MODEL_SEED = 1234
@@ -130,8 +130,10 @@ def train(cfg):
             output = model(data)
             loss = loss_fn(output, target)
 
-            # 2. backward + clipping
-            loss.backward()
+            # 2. backward + clipping (wgrad_backward_context is a no-op
+            # unless VITFSDP_NATIVE_WGRAD=2 routes dW to csrc/wgemm.hip)
+            with wgrad_backward_context():
+                loss.backward()
             if not cfg.run_without_fsdp:
                 # clip on the FULL (not per-shard) gradient norm — the
                 # shards partition the full gradient, so one scalar
