@@ -352,15 +352,17 @@ def test_native_linear_grads(monkeypatch):
                                    err_msg=name)
 
 
-def test_wgrad_dispatch_mode_real_kernel():
+def test_wgrad_dispatch_mode_real_kernel(monkeypatch):
     """NativeWgradMode (VITFSDP_NATIVE_WGRAD=2 path) on GPU: the
     dispatcher interception routes the AddmmBackward dW mm to
     csrc/wgemm.hip and matches stock autograd, with the stock addmm
     node left in place (see tests/test_native_wgrad_dispatch.py for
     the early-stop rationale)."""
     from vit_10b_fsdp_example_amd.ops import NativeWgradMode
+    import vit_10b_fsdp_example_amd.ops.linear as linear_mod
     import torch.nn as nn
 
+    monkeypatch.setattr(linear_mod, "_MIN_TILES", 0)  # small test shape
     torch.manual_seed(7)
     K, IN, OUT = 2048, 512, 768  # passes the 256x256x64 gate
     lin = nn.Linear(IN, OUT).to(_dev(), torch.bfloat16)

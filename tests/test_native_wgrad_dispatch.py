@@ -21,7 +21,13 @@ def _handler(a, b):
     return (a.t().double() @ b.double()).to(b.dtype)
 
 
-def test_pattern_matcher():
+def _no_min_tiles(monkeypatch):
+    import vit_10b_fsdp_example_amd.ops.linear as linear_mod
+    monkeypatch.setattr(linear_mod, "_MIN_TILES", 0)
+
+
+def test_pattern_matcher(monkeypatch):
+    _no_min_tiles(monkeypatch)
     x = torch.randn(K, IN)
     dy = torch.randn(K, OUT)
     assert _is_wgrad_mm(x.t(), dy)          # the AddmmBackward dW shape
@@ -30,7 +36,8 @@ def test_pattern_matcher():
     assert not _is_wgrad_mm(torch.randn(K, 100).t(), dy)  # gate: 100%256
 
 
-def test_intercepts_linear_wgrad_and_matches_stock():
+def test_intercepts_linear_wgrad_and_matches_stock(monkeypatch):
+    _no_min_tiles(monkeypatch)
     torch.manual_seed(0)
     lin = torch.nn.Linear(IN, OUT)
     x = torch.randn(K, IN, requires_grad=True)
@@ -62,10 +69,11 @@ class _CountMM(TorchDispatchMode):
         return func(*args, **(kwargs or {}))
 
 
-def test_mode_preserves_checkpoint_early_stop():
+def test_mode_preserves_checkpoint_early_stop(monkeypatch):
     """The whole point of mode "2": under the dispatch mode the stock
     addmm nodes remain, so the last recompute GEMM is still skipped —
     total backward GEMM count stays 5 (vs 6 for the Function path)."""
+    _no_min_tiles(monkeypatch)
     torch.manual_seed(0)
     w1 = torch.randn(256, 256, requires_grad=True)
     b1 = torch.randn(256, requires_grad=True)

@@ -39,10 +39,19 @@ Measurement history (all on MI355X, receipts in profiles/PROFILES.md):
     +128 forward GEMMs (+657 ms) per 4 profiled steps, ~164 ms/step —
     which is the previously "unexplained" part of the A/B gap.  The
     rest of the gap was the since-removed bias-fused wgemm variant.
-The native path is therefore OPT-IN (VITFSDP_NATIVE_WGRAD=1): the GEMM
-core is at parity, but any always-on integration must keep the stock
-autograd node for the forward (wgrad hooked below autograd, or the
-checkpoint region restructured) instead of substituting a Function.
+The corrected integration is VITFSDP_NATIVE_WGRAD=2 (NativeWgradMode
+below): dispatcher-level interception that keeps the stock addmm node.
+Measured on MI355X (ViT-Large bs=128, within-box where noted):
+  * mode 2 ungated: 1040 -> 702 img/s — the kernel has no split-K, so
+    ViT-Large dW grids (16-64 workgroups) cannot fill 256 CUs;
+  * mode 2 with the _MIN_TILES>=256 gate (nothing routed at Large):
+    117.4 ms/step vs 123.1 baseline (different boxes, within the ±5%
+    box variance) — the TorchDispatchMode python overhead itself is
+    NOT measurable: backward is GPU-bound and dispatch overlaps with
+    queued kernels.
+Both modes stay OPT-IN pending a ViT-10B within-box A/B (qkv/proj pass
+the gate there at 1200/400 workgroups); per-layer parity says expect
+neutral-to-small effect (ROADMAP item 4).
 """
 
 import contextlib
@@ -106,14 +115,23 @@ class _NativeLinearFn(torch.autograd.Function):
         return dx, dw, db
 
 
+# The kernel has no split-K: its grid is (m/256)*(n/256) workgroups, so
+# small dW matrices cannot fill 256 CUs (ViT-Large proj dW is 16
+# workgroups and measured 1040 -> 702 img/s when routed native).  10B
+# qkv/proj are 1200/400 workgroups.  Env override for experiments.
+_MIN_TILES = int(os.environ.get("VITFSDP_WGRAD_MIN_TILES", "256"))
+
+
 def _wgrad_mm_shapes_ok(m, n, k):
     """Shape gate for the native wgrad kernel: the 256x256x64 tile must
-    divide evenly, and the measured ViT-10B crossover keeps the very
+    divide evenly, the grid must be large enough to fill the chip
+    (_MIN_TILES), and the measured ViT-10B crossover keeps the very
     wide/tall MLP shapes (fc1/fc2, 20480-wide) on hipBLASLt while qkv
     and proj go native.  Symmetric bound because the dispatcher sees dW
     in whichever orientation AddmmBackward chose."""
     return (k % 64 == 0 and m % 256 == 0 and n % 256 == 0
-            and m <= 16384 and n <= 16384)
+            and m <= 16384 and n <= 16384
+            and (m // 256) * (n // 256) >= _MIN_TILES)
 
 
 def _is_wgrad_mm(at, b):
