@@ -41,3 +41,38 @@ def test_train_e2e_tiny(tmp_path, capsys):
     assert "training completed" in out
     ckpts = glob.glob(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))
     assert len(ckpts) == 1
+
+
+def test_train_resume_epoch(tmp_path, capsys):
+    """--resume_epoch N: train() loads epoch_N_rank_r.ckpt and continues
+    at epoch N+1 (reference resume semantics, run_vit_training.py)."""
+    from vit_10b_fsdp_example_amd.train import main
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    base = [
+        "--fake_data", "--image_size", "16", "--patch_size", "4",
+        "--embed_dim", "32", "--num_heads", "2", "--num_blocks", "2",
+        "--num_classes", "10", "--batch_size", "4", "--num_workers", "0",
+        "--ckpt_epoch_interval", "1", "--test_epoch_interval", "99",
+        "--log_step_interval", "1", "--warmup_steps", "2",
+        "--max_steps_per_epoch", "2", "--ckpt_dir", str(tmp_path),
+    ]
+    import vit_10b_fsdp_example_amd.data.datasets as ds
+
+    orig_train, orig_val = ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN
+    ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN = 8, 4
+    try:
+        CommContext.reset()
+        main(parse_args(base + ["--num_epochs", "1"]))
+        assert os.path.exists(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))
+        capsys.readouterr()
+
+        CommContext.reset()
+        main(parse_args(base + ["--num_epochs", "2", "--resume_epoch", "1"]))
+    finally:
+        ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN = orig_train, orig_val
+
+    out = capsys.readouterr().out
+    assert "starting epoch 2" in out
+    assert "starting epoch 1" not in out  # resumed, not restarted
+    assert os.path.exists(os.path.join(str(tmp_path), "epoch_2_rank_0.ckpt"))
