@@ -119,3 +119,22 @@ def test_ws4_fsdp_matches_single_rank():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def _run_trajectory_prefetch2(rank, world_size, extra):
+    # widen the async-gather window (VITFSDP_PREFETCH_DEPTH=2): the
+    # collective ORDER must stay uniform across ranks and numerics
+    # unchanged — this is the 8-GPU overlap knob (ROADMAP item 1)
+    import vit_10b_fsdp_example_amd.parallel.fsdp as fsdp_mod
+
+    fsdp_mod._PREFETCH_DEPTH = 2
+    return _run_trajectory(rank, world_size, extra)
+
+
+def test_ws2_fsdp_prefetch_depth2():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory_prefetch2, world_size=2,
+                               args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)

@@ -37,6 +37,8 @@ reductions and installs fp32 mean gradients on the master shards, so
 `optimizer.step()` needs no knowledge of any of this.
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -45,6 +47,11 @@ from .comm import CommContext
 
 # Modules currently executing an FSDP forward (root detection).
 _EXEC_STACK = []
+
+# How many units ahead to issue async param gathers (both directions).
+# 1 = classic one-ahead overlap; raise for 8-GPU runs where a gather
+# outlasts a block's compute (see _prefetch_neighbor).
+_PREFETCH_DEPTH = max(1, int(os.environ.get("VITFSDP_PREFETCH_DEPTH", "1")))
 
 
 def _free_storage(t):
@@ -304,15 +311,23 @@ class FullyShardedDataParallel(nn.Module):
         return out
 
     def _prefetch_neighbor(self, root, direction):
+        """Issue async gathers for the next _PREFETCH_DEPTH units in
+        execution order (+1 = forward, -1 = backward).  Depth 1 hides
+        one unit's all-gather behind one unit's compute; on 8 GPUs the
+        gather may take longer than a block (xGMI ring latency), so
+        VITFSDP_PREFETCH_DEPTH widens the window at the cost of keeping
+        that many extra units' full params resident (~600 MB each for
+        ViT-10B)."""
         if not self.prefetch or root._fwd_order is None:
             return
         try:
             i = root._fwd_order.index(self)
         except ValueError:
             return
-        j = i + direction
-        if 0 <= j < len(root._fwd_order):
-            root._fwd_order[j]._prefetch_gather()
+        for step in range(1, _PREFETCH_DEPTH + 1):
+            j = i + direction * step
+            if 0 <= j < len(root._fwd_order):
+                root._fwd_order[j]._prefetch_gather()
 
     def _register_pre_backward(self, out):
         tensors = out if isinstance(out, (tuple, list)) else (out,)
