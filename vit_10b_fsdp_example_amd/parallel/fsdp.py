@@ -143,7 +143,12 @@ class FullyShardedDataParallel(nn.Module):
             shard = flat.narrow(
                 0, rank * self._shard_numel, self._shard_numel
             ).clone()
-            if self.device.type == "cuda":
+            # VITFSDP_NO_PIN=1 keeps host shards pageable: slower H2D
+            # staging, but the safe mode for single-GPU 60B-class runs
+            # where pinning ~230 GB of host memory can OOM the box
+            # (at ws=8 the per-rank ~29 GB pinned is unproblematic)
+            if (self.device.type == "cuda"
+                    and os.environ.get("VITFSDP_NO_PIN", "0") != "1"):
                 shard = shard.pin_memory()
         else:
             shard = (
