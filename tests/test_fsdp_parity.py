@@ -138,3 +138,31 @@ def test_ws2_fsdp_prefetch_depth2():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def _run_trajectory_p2p_gather(rank, world_size, extra):
+    # direct one-shot all-gather (batched isend/irecv into full-buffer
+    # slices) instead of the library collective — the xGMI full-mesh
+    # algorithm candidate from SURVEY §5 (VITFSDP_AG_ALGO=p2p)
+    import vit_10b_fsdp_example_amd.parallel.comm as comm_mod
+
+    comm_mod._AG_ALGO = "p2p"
+    return _run_trajectory(rank, world_size, extra)
+
+
+def test_ws2_fsdp_p2p_allgather():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory_p2p_gather, world_size=2,
+                               args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_ws4_fsdp_p2p_allgather():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory_p2p_gather, world_size=4,
+                               args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)

@@ -17,6 +17,8 @@ Design notes (MI355X / RCCL over xGMI):
     very same FSDP control flow.
 """
 
+import os
+
 import torch
 import torch.distributed as dist
 
@@ -26,7 +28,33 @@ class _NoopWork:
         return True
 
 
+class _MultiWork:
+    """Aggregate of several async work handles (one-shot P2P gathers)."""
+
+    def __init__(self, works):
+        self._works = works
+
+    def wait(self):
+        for w in self._works:
+            w.wait()
+        return True
+
+
 _NOOP = _NoopWork()
+
+# All-gather algorithm (SURVEY.md §5 "Distributed communication backend"):
+#   "allgather" — the RCCL collective (default; RCCL picks its own
+#                 multi-link algorithm for large messages)
+#   "p2p"       — direct one-shot: every rank sends its shard to all
+#                 ws-1 peers and receives each peer's shard straight
+#                 into its slice of the full buffer, in one batched
+#                 group.  On the xGMI full mesh this drives all 7 links
+#                 simultaneously instead of a per-link-bound ring, with
+#                 zero staging copies (the recv targets are contiguous
+#                 slices of the destination).
+# Correctness is backend-independent (gloo parity tests cover p2p);
+# which wins on real xGMI is a round-2 A/B (ROADMAP item 1).
+_AG_ALGO = os.environ.get("VITFSDP_AG_ALGO", "allgather")
 
 
 def _backend_is_gloo(group):
@@ -74,6 +102,26 @@ class CommContext:
 
     # -- param all-gather ---------------------------------------------------
 
+    def _all_gather_p2p(self, full, shard, async_op):
+        """Direct one-shot all-gather: batched isend/irecv to/from every
+        peer, receiving straight into `full`'s per-rank slices."""
+        n = shard.numel()
+        full.narrow(0, self.rank * n, n).copy_(shard)
+        ops = []
+        for peer in range(self.world_size):
+            if peer == self.rank:
+                continue
+            ops.append(dist.P2POp(dist.isend, shard, peer,
+                                  group=self.gather_group))
+            ops.append(dist.P2POp(dist.irecv, full.narrow(0, peer * n, n),
+                                  peer, group=self.gather_group))
+        works = dist.batch_isend_irecv(ops)
+        if async_op:
+            return _MultiWork(works)
+        for w in works:
+            w.wait()
+        return _NOOP
+
     def all_gather_into(self, full, shard, async_op=False):
         """Gather each rank's `shard` into `full` (full.numel == ws * shard.numel).
 
@@ -81,6 +129,8 @@ class CommContext:
         if self.world_size == 1:
             full.copy_(shard)
             return _NOOP
+        if _AG_ALGO == "p2p":
+            return self._all_gather_p2p(full, shard, async_op)
         if _backend_is_gloo(self.gather_group):
             try:
                 dist.all_gather_into_tensor(full, shard, group=self.gather_group)
