@@ -166,3 +166,30 @@ def test_ws4_fsdp_p2p_allgather():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def _run_trajectory_p2p_both(rank, world_size, extra):
+    # both collectives on the one-shot P2P algorithms at once
+    import vit_10b_fsdp_example_amd.parallel.comm as comm_mod
+
+    comm_mod._AG_ALGO = "p2p"
+    comm_mod._RS_ALGO = "p2p"
+    return _run_trajectory(rank, world_size, extra)
+
+
+def test_ws2_fsdp_p2p_reduce_scatter():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory_p2p_both, world_size=2,
+                               args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_ws4_fsdp_p2p_both_algos():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory_p2p_both, world_size=4,
+                               args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)

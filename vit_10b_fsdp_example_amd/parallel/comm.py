@@ -56,6 +56,29 @@ _NOOP = _NoopWork()
 # which wins on real xGMI is a round-2 A/B (ROADMAP item 1).
 _AG_ALGO = os.environ.get("VITFSDP_AG_ALGO", "allgather")
 
+# Reduce-scatter algorithm: "reducescatter" (RCCL collective, default)
+# or "p2p" — direct scatter + local reduction: every rank sends slice p
+# of its full gradient to peer p, receives ws-1 peer slices, and sums
+# them with its own slice at wait() time.
+_RS_ALGO = os.environ.get("VITFSDP_RS_ALGO", "reducescatter")
+
+
+class _DeferredWork:
+    """Work handle that runs a local finalizer after the async sends/
+    receives complete (one-shot P2P reduce-scatter)."""
+
+    def __init__(self, works, finish):
+        self._works = works
+        self._finish = finish
+
+    def wait(self):
+        for w in self._works:
+            w.wait()
+        if self._finish is not None:
+            self._finish()
+            self._finish = None
+        return True
+
 
 def _backend_is_gloo(group):
     if not dist.is_initialized():
@@ -145,12 +168,44 @@ class CommContext:
 
     # -- grad reduce-scatter ------------------------------------------------
 
+    def _reduce_scatter_p2p(self, out_shard, full, async_op):
+        """Direct scatter + local reduce: slice p of `full` goes to peer
+        p; the ws-1 received slices are summed with our own slice when
+        the handle is waited.  The caller must keep `full` alive until
+        wait() (the FSDP engine frees the grad buffer only after
+        finalization)."""
+        n = out_shard.numel()
+        stage = torch.empty((self.world_size - 1) * n, dtype=full.dtype,
+                            device=full.device)
+        ops, slot = [], 0
+        for peer in range(self.world_size):
+            if peer == self.rank:
+                continue
+            ops.append(dist.P2POp(dist.isend, full.narrow(0, peer * n, n),
+                                  peer, group=self.reduce_group))
+            ops.append(dist.P2POp(dist.irecv, stage.narrow(0, slot * n, n),
+                                  peer, group=self.reduce_group))
+            slot += 1
+        works = dist.batch_isend_irecv(ops)
+
+        def finish():
+            out_shard.copy_(full.narrow(0, self.rank * n, n))
+            out_shard.add_(stage.view(self.world_size - 1, n).sum(0))
+
+        work = _DeferredWork(works, finish)
+        if async_op:
+            return work
+        work.wait()
+        return _NOOP
+
     def reduce_scatter_into(self, out_shard, full, async_op=False):
         """Sum-reduce `full` across ranks, scattering shard `rank` into
         `out_shard`.  Caller divides by world size (mean semantics)."""
         if self.world_size == 1:
             out_shard.copy_(full)
             return _NOOP
+        if _RS_ALGO == "p2p":
+            return self._reduce_scatter_p2p(out_shard, full, async_op)
         if _backend_is_gloo(self.reduce_group):
             dist.all_reduce(full, group=self.reduce_group)
             n = out_shard.numel()
