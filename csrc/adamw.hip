@@ -10,8 +10,10 @@
 // over each tensor in turn with float4 (16 B) accesses.  Purely
 // memory-bound; one launch per <=32 tensors.
 
+#ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
+#endif
 
 #include "common.h"
 
@@ -149,6 +151,7 @@ int grid_for(long total_elems) {
 
 }  // namespace
 
+#ifndef VITFSDP_KERNELS_ONLY
 void fused_adamw(std::vector<torch::Tensor> params,
                  std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> exp_avgs,
@@ -259,3 +262,5 @@ void multi_tensor_scale_tensor(std::vector<torch::Tensor> tensors,
               "factor must be a CUDA fp32 scalar tensor");
   mt_scale_impl(tensors, factor.data_ptr<float>(), 1.f);
 }
+
+#endif  // VITFSDP_KERNELS_ONLY

@@ -4,8 +4,10 @@
 // run_vit_training.py:229,262).  Shapes are small ([batch, 1000]); one
 // 256-thread block per row, loss accumulated with one fp32 atomic.
 
+#ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
+#endif
 
 #include "common.h"
 
@@ -62,6 +64,7 @@ __global__ void ce_bwd_kernel(const unsigned short* __restrict__ logits,
 
 }  // namespace
 
+#ifndef VITFSDP_KERNELS_ONLY
 std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
                                              torch::Tensor target) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2);
@@ -97,3 +100,5 @@ torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
   HIP_CHECK_LAST();
   return dlogits;
 }
+
+#endif  // VITFSDP_KERNELS_ONLY

@@ -8,8 +8,10 @@
 // inside the 4 MB per-XCD L2), which keeps register pressure flat across
 // any D instead of caching the row in registers.
 
+#ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
+#endif
 
 #include "common.h"
 
@@ -251,6 +253,7 @@ __global__ void ln_add_bwd_dx_kernel(const ushort8_t* __restrict__ dy,
 
 }  // namespace
 
+#ifndef VITFSDP_KERNELS_ONLY
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
                                          torch::Tensor b, double eps) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
@@ -395,3 +398,5 @@ std::vector<torch::Tensor> layernorm_add_bwd(torch::Tensor dy,
   HIP_CHECK_LAST();
   return {dx1, dx2, dw, db};
 }
+
+#endif  // VITFSDP_KERNELS_ONLY

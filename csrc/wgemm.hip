@@ -25,8 +25,10 @@
 // before this step's MFMA phases.
 
 #ifndef VITFSDP_KERNELS_ONLY
+#ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
+#endif
 #endif
 
 #include "common.h"
@@ -287,6 +289,7 @@ template __global__ void wgemm_atb_kernel<false>(const short*, const short*,
 }  // namespace
 
 #ifndef VITFSDP_KERNELS_ONLY
+#ifndef VITFSDP_KERNELS_ONLY
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias) {
   TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
@@ -334,3 +337,5 @@ std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
   return {c};
 }
 #endif
+
+#endif  // VITFSDP_KERNELS_ONLY
