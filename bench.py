@@ -38,6 +38,7 @@ def parse():
     p.add_argument("--shard_on_cpu", action="store_true")
     p.add_argument("--no_grad_ckpt", action="store_false", dest="grad_ckpt")
     p.add_argument("--grad_ckpt_blocks", type=int, default=-1)
+    p.add_argument("--fuse_residual", action="store_true")
     return p.parse_args()
 
 
@@ -76,6 +77,7 @@ def main():
         "--batch_size", str(args.per_gpu_batch * world),
     ] + (["--shard_on_cpu"] if args.shard_on_cpu else [])
       + ([] if args.grad_ckpt else ["--no_grad_ckpt"])
+      + (["--fuse_residual"] if args.fuse_residual else [])
       + (["--grad_ckpt_blocks", str(args.grad_ckpt_blocks)]
          if args.grad_ckpt_blocks >= 0 else []))
 

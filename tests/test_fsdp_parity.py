@@ -193,3 +193,28 @@ def test_ws4_fsdp_p2p_both_algos():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_ws1_fuse_residual_matches_baseline():
+    """--fuse_residual (deferred-residual block interface, every add
+    fused into the next LN) is numerically identical to the default
+    interface — same sums, same rounding points in fp32."""
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"],
+                                          dropout=True)
+    for extra in (["--fuse_residual"],
+                  ["--fuse_residual", "--no_grad_ckpt"],
+                  ["--fuse_residual", "--no_reshard_after_forward"]):
+        l, g = _single_rank_reference(extra, dropout=True)
+        np.testing.assert_allclose(l, ref_l, rtol=1e-5, atol=1e-6,
+                                   err_msg=str(extra))
+        np.testing.assert_allclose(g, ref_g, rtol=1e-5, atol=1e-6,
+                                   err_msg=str(extra))
+
+
+def test_ws2_fuse_residual_matches_single_rank():
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory, world_size=2,
+                               args=(["--fuse_residual"],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)

@@ -73,6 +73,13 @@ def build_arg_parser():
         "changes.",
     )
     parser.add_argument(
+        "--fuse_residual", action="store_true", dest="fuse_residual",
+        help="deferred-residual block interface: blocks exchange "
+        "(hidden, stream) pairs and every residual add is fused into "
+        "the following LayerNorm kernel (identical math/rounding; no "
+        "standalone elementwise add kernels remain)",
+    )
+    parser.add_argument(
         "--profile", action="store_true", dest="profile",
         help="profile a few early steps with torch.profiler (CPU+GPU "
         "kernel timeline) and write a chrome trace + a top-kernel table "

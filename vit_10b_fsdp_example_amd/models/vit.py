@@ -137,11 +137,24 @@ class Mlp(nn.Module):
 class Block(nn.Module):
     """Pre-LN transformer block: x + Attn(LN1(x)); x + MLP(LN2(x))
     (timm Block as configured at reference run_vit_training.py:134-141;
-    drop_path unused there, so none here)."""
+    drop_path unused there, so none here).
+
+    Two interfaces, identical math:
+      * default: tensor -> tensor, with the attention residual add fused
+        into norm2 (one ln_add kernel) and the MLP residual as a plain
+        add;
+      * deferred residual (Megatron-style, --fuse_residual): input and
+        output are (hidden, stream) pairs and the MLP residual add is
+        fused into the NEXT block's norm1 — no standalone elementwise
+        add kernels remain anywhere in the block stack.  The bf16
+        rounding of every sum matches the plain add exactly (ln_add
+        rounds the fp32 sum to bf16 the same way torch's add does).
+    """
 
     def __init__(self, dim, num_heads, mlp_ratio=4.0, qkv_bias=True,
-                 drop=0.0, attn_drop=0.0):
+                 drop=0.0, attn_drop=0.0, deferred_residual=False):
         super().__init__()
+        self.deferred_residual = deferred_residual
         self.norm1 = LayerNorm(dim, eps=1e-6)
         self.attn = Attention(
             dim, num_heads, qkv_bias=qkv_bias, attn_drop=attn_drop, proj_drop=drop
@@ -150,6 +163,8 @@ class Block(nn.Module):
         self.mlp = Mlp(dim, int(dim * mlp_ratio), drop=drop)
 
     def forward(self, x):
+        if self.deferred_residual:
+            return self._forward_deferred(x)
         attn_out = self.attn(self.norm1(x))
         # fused residual add + norm2 (one kernel for x+attn_out and its LN)
         x, normed = fused_add_layer_norm(
@@ -157,6 +172,22 @@ class Block(nn.Module):
         )
         x = x + self.mlp(normed)
         return x
+
+    def _forward_deferred(self, x):
+        if isinstance(x, (tuple, list)):
+            hidden, stream = x
+            # previous block's MLP residual add, fused into our norm1
+            s0, y0 = fused_add_layer_norm(
+                stream, hidden, self.norm1.weight, self.norm1.bias,
+                self.norm1.eps,
+            )
+        else:  # first block: nothing pending
+            s0, y0 = x, self.norm1(x)
+        attn_out = self.attn(y0)
+        s1, y1 = fused_add_layer_norm(
+            s0, attn_out, self.norm2.weight, self.norm2.bias, self.norm2.eps
+        )
+        return self.mlp(y1), s1
 
 
 class FSDPViTModel(nn.Module):
@@ -179,8 +210,10 @@ class FSDPViTModel(nn.Module):
         num_classes,
         grad_ckpt_wrap,
         fsdp_wrap,
+        fuse_residual=False,
     ):
         super().__init__()
+        self.fuse_residual = fuse_residual
 
         self.patch_embed = PatchEmbed(
             img_size=image_size, patch_size=patch_size, in_chans=3,
@@ -201,6 +234,7 @@ class FSDPViTModel(nn.Module):
                 qkv_bias=True,
                 drop=mlp_dropout,
                 attn_drop=att_dropout,
+                deferred_residual=fuse_residual,
             )
             # init BEFORE wrapping: FSDP shards the weights at wrap time
             init_vit_weights(block)
@@ -220,8 +254,18 @@ class FSDPViTModel(nn.Module):
         x = self.patch_embed(image) + self.pos_embed
         x = self.pos_drop(x)
         x = self.blocks(x)
+        if self.fuse_residual:
+            # last block's pending MLP residual add fuses into the
+            # final LayerNorm
+            hidden, stream = x
+            _, normed = fused_add_layer_norm(
+                stream, hidden, self.norm.weight, self.norm.bias,
+                self.norm.eps,
+            )
+        else:
+            normed = self.norm(x)
         # mean pooling over the patch sequence instead of a [CLS] token
-        logits = self.head(torch.mean(self.norm(x), dim=1))
+        logits = self.head(torch.mean(normed, dim=1))
         return logits
 
 
@@ -298,6 +342,7 @@ def build_fsdp_vit_model(cfg, device, compute_dtype=torch.float32):
         num_classes=cfg.num_classes,
         grad_ckpt_wrap=grad_ckpt_wrap,
         fsdp_wrap=fsdp_wrap,
+        fuse_residual=getattr(cfg, "fuse_residual", False),
     )
     # root wrap without grad-ckpt (reference run_vit_training.py:197-199)
     model = fsdp_wrap(model)

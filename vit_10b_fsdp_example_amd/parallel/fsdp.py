@@ -335,12 +335,18 @@ class FullyShardedDataParallel(nn.Module):
                 root._fwd_order[j]._prefetch_gather()
 
     def _register_pre_backward(self, out):
+        # hook EVERY grad-carrying output: with a multi-tensor output
+        # (e.g. the deferred-residual (hidden, stream) block interface)
+        # the engine may route either tensor's gradient into this unit
+        # first, and params must be re-gathered before any of them.
+        # _pre_bwd_done makes the extra firings no-ops.
         tensors = out if isinstance(out, (tuple, list)) else (out,)
+        any_grad = False
         for t in tensors:
             if torch.is_tensor(t) and t.requires_grad:
                 t.register_hook(self._pre_backward_hook)
-                return True
-        return False
+                any_grad = True
+        return any_grad
 
     # ------------------------------------------------------------------
     # backward
