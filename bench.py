@@ -167,6 +167,7 @@ def main():
                 "parallelism": f"fsdp{world}" + ("+cpu_shard" if args.shard_on_cpu else ""),
                 "grad_ckpt": cfg.grad_ckpt,
                 "grad_ckpt_blocks": args.grad_ckpt_blocks,
+                "fuse_residual": args.fuse_residual,
                 "final_loss": float(loss.item()) if loss is not None else None,
             },
         }

@@ -1,12 +1,14 @@
 """hipBLASLt/rocBLAS GEMM algorithm selection (PyTorch TunableOp).
 
-The repo ships a pre-tuned solution table for the ViT-10B GEMM shapes on
-gfx950 (tuned/tunableop_gfx950.csv, produced offline on an MI355X).
-Loading it is read-only: tuning itself stays disabled at runtime, ops
-without an entry fall back to the default heuristic, numerics are
-unchanged (same library kernels, different algorithm choice).
-
-Disable with VITFSDP_TUNABLEOP=0.
+STRICTLY OPT-IN (VITFSDP_TUNABLEOP=1): TunableOp — both live tuning and
+read-only table loading — crashes this ROCm 7.2 / torch 2.10 stack on
+gfx950 (see enable_tunableop below), so the default path never touches
+it.  When opted in, the pre-tuned table for the ViT-10B GEMM shapes
+(tuned/tunableop_gfx950.csv, produced offline on an MI355X) is loaded
+read-only: ops without an entry fall back to the default heuristic and
+numerics are unchanged (same library kernels, different algorithm
+choice).  The maintained alternative is the standalone
+csrc/tools/hipblaslt_search.cpp enumerator (ROADMAP item 5).
 """
 
 import os
