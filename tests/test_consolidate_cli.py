@@ -27,3 +27,22 @@ def test_cli_subprocess(tmp_path):
     full = torch.load(out, map_location="cpu", weights_only=False)
     assert "model" in full
     assert any(k.endswith("attn.qkv.weight") for k in full["model"])
+
+
+def test_missing_rank_file_fails_loudly(tmp_path):
+    """Consolidating with a rank file absent must abort, not silently
+    produce a half-assembled model."""
+    import pytest
+
+    ckpt_dir = str(tmp_path)
+    run_multiprocess(_train_save, world_size=2, args=(ckpt_dir,))
+    os.remove(os.path.join(ckpt_dir, "epoch_1_rank_1.ckpt"))
+
+    from vit_10b_fsdp_example_amd.consolidate_sharded_ckpts import (
+        consolidate_files,
+    )
+
+    with pytest.raises(AssertionError, match="rank files"):
+        consolidate_files(os.path.join(ckpt_dir, "epoch_1_rank_"), ".ckpt",
+                          os.path.join(ckpt_dir, "full.ckpt"))
+    assert not os.path.exists(os.path.join(ckpt_dir, "full.ckpt"))
