@@ -1,0 +1,53 @@
+"""bench.py driver contract: `python bench.py --gpus N --steps K
+--warmup W` must print ONE JSON line from rank 0 with the agreed
+fields/types.  The round-end BENCH/SCALE runs depend on this shape —
+this test keeps refactors from breaking it."""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract():
+    res = subprocess.run(
+        [sys.executable, "bench.py", "--model", "vit-tiny",
+         "--per_gpu_batch", "2", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, cwd=REPO, timeout=600,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    json_lines = [ln for ln in res.stdout.splitlines()
+                  if ln.startswith("{") and ln.endswith("}")]
+    assert len(json_lines) == 1, res.stdout[-2000:]
+    out = json.loads(json_lines[0])
+
+    assert isinstance(out["metric"], str) and "images/sec" in out["metric"]
+    assert isinstance(out["value"], (int, float)) and out["value"] > 0
+    assert out["unit"] == "images/sec"
+    assert out["n_gpus"] == 1
+    assert out["steps"] == 1
+    assert out["warmup"] == 0
+    assert isinstance(out["ms_per_step"], (int, float)) and out["ms_per_step"] > 0
+    assert out["higher_is_better"] is True
+    assert out["scaling"] == "weak"
+    assert out["vs_baseline"] is None  # reference publishes no number
+    assert out["dtype"] in ("bf16", "fp32")
+    assert out["data"] == "synthetic"
+    cfg = out["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg, key
+    assert cfg["global_batch"] == 2
+    # value must be consistent with ms_per_step at N=1
+    implied = 1000.0 / out["ms_per_step"] * cfg["global_batch"]
+    assert abs(implied - out["value"]) / out["value"] < 0.05
+
+
+def test_bench_default_metric_names_baseline_config():
+    """With no flags the metric string must name the BASELINE.json
+    headline (ViT-10B bs=1024 224px) — checked statically so we don't
+    build a 10B model on CPU."""
+    src = open(os.path.join(REPO, "bench.py")).read()
+    assert "images/sec (whole node) for ViT-10B bs=1024 224px --fake_data" in src
+    assert '"vit10b"' in src
