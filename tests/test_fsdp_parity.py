@@ -218,3 +218,14 @@ def test_ws2_fuse_residual_matches_single_rank():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_ws8_fsdp_matches_single_rank():
+    """8-rank FSDP (gloo) — the exact rank count of the driver's
+    round-end scaling run; exercises the full 8-way collective schedule
+    and shard padding (batch 8 -> 1 image/rank)."""
+    ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    results = run_multiprocess(_run_trajectory, world_size=8, args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
