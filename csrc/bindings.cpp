@@ -55,6 +55,7 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
@@ -83,4 +84,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("wgrad_gemm", &wgrad_gemm,
         "C = A^T B weight-gradient GEMM (bf16, tr16 transpose reads)");
+  m.def("lt_gemm", &lt_gemm,
+        "row-major bf16 GEMM via hipblaslt-ext with an explicit "
+        "algorithm index (-1 = library heuristic); offline-search "
+        "apply path, see csrc/tools/hipblaslt_search.cpp");
 }

@@ -18,12 +18,14 @@ ext = CUDAExtension(
     name="vit_10b_fsdp_example_amd._C",
     sources=[
         "csrc/bindings.cpp",
+        "csrc/ltgemm.cpp",
         "csrc/layernorm.hip",
         "csrc/adamw.hip",
         "csrc/cross_entropy.hip",
         "csrc/fmha.hip",
         "csrc/wgemm.hip",
     ],
+    libraries=["hipblaslt"],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": [
