@@ -55,7 +55,8 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
-torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index);
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
+                      c10::optional<torch::Tensor> bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
@@ -85,7 +86,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_gemm", &wgrad_gemm,
         "C = A^T B weight-gradient GEMM (bf16, tr16 transpose reads)");
   m.def("lt_gemm", &lt_gemm,
-        "row-major bf16 GEMM via hipblaslt-ext with an explicit "
-        "algorithm index (-1 = library heuristic); offline-search "
-        "apply path, see csrc/tools/hipblaslt_search.cpp");
+        "row-major bf16 GEMM (+optional fused bias epilogue) via "
+        "hipblaslt-ext with an explicit algorithm index (-1 = library "
+        "heuristic); offline-search apply path, see "
+        "csrc/tools/hipblaslt_search.cpp",
+        py::arg("a"), py::arg("b"), py::arg("algo_index") = -1,
+        py::arg("bias") = py::none());
 }

@@ -56,7 +56,8 @@ constexpr size_t kMaxWorkspace = 128u << 20;
 
 }  // namespace
 
-torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index) {
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
+                      c10::optional<torch::Tensor> bias) {
   TORCH_CHECK(a.is_cuda() && b.is_cuda(), "lt_gemm: CUDA tensors required");
   TORCH_CHECK(a.dtype() == torch::kBFloat16 && b.dtype() == torch::kBFloat16,
               "lt_gemm: bf16 only");
@@ -76,6 +77,18 @@ torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index) {
   float alpha = 1.0f, beta = 0.0f;
   hipblaslt_ext::GemmEpilogue epilogue;
   hipblaslt_ext::GemmInputs inputs;
+  torch::Tensor bias_c;
+  if (bias.has_value()) {
+    // per-output-feature bias: length N = rows of the column-major dual,
+    // which is exactly hipBLASLt's bias-vector broadcast
+    bias_c = bias->contiguous();
+    TORCH_CHECK(bias_c.is_cuda() && bias_c.dtype() == torch::kBFloat16 &&
+                    bias_c.numel() == N,
+                "lt_gemm: bias must be bf16 CUDA of length N");
+    epilogue.setMode(HIPBLASLT_EPILOGUE_BIAS);
+    epilogue.setBiasDataType(HIP_R_16BF);
+    inputs.setBias(bias_c.data_ptr());
+  }
   inputs.setA(bc.data_ptr());  // column-major dual: A <- b
   inputs.setB(ac.data_ptr());
   inputs.setC(d.data_ptr());

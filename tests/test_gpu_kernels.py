@@ -408,3 +408,9 @@ def test_lt_gemm_matches_matmul():
     scale = ref.abs().max().item()
     np.testing.assert_allclose(d.float().cpu().numpy(), ref.cpu().numpy(),
                                atol=0.02 * scale, rtol=0.05)
+    # fused bias epilogue: per-output-feature vector of length N
+    bias = torch.randn(N, device=_dev(), dtype=torch.bfloat16)
+    db = EXT.lt_gemm(a, b, -1, bias)
+    refb = ref + bias.float()
+    np.testing.assert_allclose(db.float().cpu().numpy(), refb.cpu().numpy(),
+                               atol=0.02 * scale, rtol=0.05)
