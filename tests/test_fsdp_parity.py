@@ -229,3 +229,33 @@ def test_ws8_fsdp_matches_single_rank():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+PAD = [
+    "--fake_data", "--image_size", "32", "--patch_size", "4",
+    "--embed_dim", "20", "--num_heads", "4", "--num_blocks", "2",
+    "--num_classes", "13", "--batch_size", "8", "--num_workers", "0",
+]
+
+
+def _run_trajectory_pad(rank, world_size, extra):
+    # root-unit total is 2573 params (odd): at ws=2 the flat shard is
+    # PADDED, exercising the zero-tail in gather and reduce-scatter
+    global TINY
+    saved, TINY[:] = TINY[:], PAD
+    try:
+        return _run_trajectory(rank, world_size, extra)
+    finally:
+        TINY[:] = saved
+
+
+def test_ws2_fsdp_padded_shards():
+    saved, TINY[:] = TINY[:], PAD
+    try:
+        ref_l, ref_g = _single_rank_reference(["--run_without_fsdp"])
+    finally:
+        TINY[:] = saved
+    results = run_multiprocess(_run_trajectory_pad, world_size=2, args=([],))
+    for l, g in results:
+        np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
