@@ -5,14 +5,14 @@ Capability parity with the reference's build_datasets
   * --fake_data swaps in a zero-image dataset with ImageNet-1k lengths
     (train 1,281,167 / val 50,000) — the de-facto integration-test and
     benchmarking mode,
-  * real mode: torchvision ImageFolder with the standard ImageNet
-    train/val transforms (RandomResizedCrop+flip / Resize+CenterCrop,
-    bicubic, ImageNet mean/std),
+  * real mode: our own PIL-based ImageFolder with the standard
+    ImageNet train/val transforms (RandomResizedCrop+flip /
+    Resize+CenterCrop, bicubic, ImageNet mean/std),
   * per-rank DistributedSampler with drop_last=True on both splits
     (shuffle only for train), local batch = global batch / world size.
 
-torchvision is imported lazily: it is only needed for the real-data
-path, and the synthetic path must work on boxes without it.
+The real-data path is self-contained (PIL + torch, imagefolder.py) —
+torchvision is not installed in the target image.
 """
 
 import os
@@ -45,13 +45,15 @@ class FakeImageNetDataset(torch.utils.data.Dataset):
 
 
 def _imagefolder_datasets(cfg):
-    import torchvision
-    import torchvision.transforms as T
+    # self-contained PIL+torch ImageFolder/transforms (imagefolder.py):
+    # torchvision is not installed in the target image, and the
+    # transforms replicate its semantics (reference
+    # run_vit_training.py:40-55) directly
+    from . import imagefolder as T
 
-    interp = T.InterpolationMode.BICUBIC
     train_transform = T.Compose(
         [
-            T.RandomResizedCrop(cfg.image_size, interpolation=interp),
+            T.RandomResizedCrop(cfg.image_size),
             T.RandomHorizontalFlip(),
             T.ToTensor(),
             T.Normalize(mean=IMAGENET_MEAN, std=IMAGENET_STD),
@@ -59,18 +61,16 @@ def _imagefolder_datasets(cfg):
     )
     val_transform = T.Compose(
         [
-            T.Resize((cfg.image_size * 256) // 224, interpolation=interp),
+            T.Resize((cfg.image_size * 256) // 224),
             T.CenterCrop(cfg.image_size),
             T.ToTensor(),
             T.Normalize(mean=IMAGENET_MEAN, std=IMAGENET_STD),
         ]
     )
-    train_ds = torchvision.datasets.ImageFolder(
+    train_ds = T.ImageFolder(
         os.path.join(cfg.data_dir, "train"), train_transform
     )
-    val_ds = torchvision.datasets.ImageFolder(
-        os.path.join(cfg.data_dir, "val"), val_transform
-    )
+    val_ds = T.ImageFolder(os.path.join(cfg.data_dir, "val"), val_transform)
     return train_ds, val_ds
 
 
