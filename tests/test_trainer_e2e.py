@@ -76,3 +76,26 @@ def test_train_resume_epoch(tmp_path, capsys):
     assert "starting epoch 2" in out
     assert "starting epoch 1" not in out  # resumed, not restarted
     assert os.path.exists(os.path.join(str(tmp_path), "epoch_2_rank_0.ckpt"))
+
+
+def test_cli_entry_subprocess(tmp_path):
+    """The run_vit_training.py entry (parse -> spawn -> train) as a real
+    subprocess, inline single-process path."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = subprocess.run(
+        [sys.executable, "run_vit_training.py",
+         "--fake_data", "--image_size", "16", "--patch_size", "4",
+         "--embed_dim", "32", "--num_heads", "2", "--num_blocks", "1",
+         "--num_classes", "10", "--batch_size", "4", "--num_workers", "0",
+         "--num_epochs", "1", "--max_steps_per_epoch", "2",
+         "--warmup_steps", "1", "--test_epoch_interval", "99",
+         "--ckpt_dir", str(tmp_path)],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+        env={**os.environ, "VITFSDP_FAKE_LEN": "8"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    assert "training completed" in res.stdout
+    assert os.path.exists(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))

@@ -85,8 +85,15 @@ def build_datasets(cfg, device, compute_dtype=None):
 
     if cfg.fake_data:
         xdist.master_print("loading fake images")
-        train_dataset = FakeImageNetDataset(cfg.image_size, IMAGENET_TRAIN_LEN)
-        val_dataset = FakeImageNetDataset(cfg.image_size, IMAGENET_VAL_LEN)
+        # VITFSDP_FAKE_LEN overrides both synthetic split lengths
+        # (smoke tests / CI; 0 = the real ImageNet-1k lengths)
+        fake_len = int(os.environ.get("VITFSDP_FAKE_LEN", "0"))
+        train_dataset = FakeImageNetDataset(
+            cfg.image_size, fake_len or IMAGENET_TRAIN_LEN
+        )
+        val_dataset = FakeImageNetDataset(
+            cfg.image_size, fake_len or IMAGENET_VAL_LEN
+        )
     else:
         xdist.master_print(f"loading images from directory: {cfg.data_dir}")
         train_dataset, val_dataset = _imagefolder_datasets(cfg)
