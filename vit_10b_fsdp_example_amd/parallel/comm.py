@@ -189,8 +189,13 @@ class CommContext:
         works = dist.batch_isend_irecv(ops)
 
         def finish():
-            out_shard.copy_(full.narrow(0, self.rank * n, n))
-            out_shard.add_(stage.view(self.world_size - 1, n).sum(0))
+            # accumulate in fp32 with ONE final rounding to the comm
+            # dtype — a ring reduce-scatter rounds to bf16 at every hop
+            # (ws-1 times); the one-shot layout gets better gradient
+            # numerics at identical comm volume
+            acc = full.narrow(0, self.rank * n, n).to(torch.float32)
+            acc += stage.view(self.world_size - 1, n).to(torch.float32).sum(0)
+            out_shard.copy_(acc)
 
         work = _DeferredWork(works, finish)
         if async_op:
