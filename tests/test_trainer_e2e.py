@@ -99,3 +99,33 @@ def test_cli_entry_subprocess(tmp_path):
     assert res.returncode == 0, res.stderr[-2000:]
     assert "training completed" in res.stdout
     assert os.path.exists(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))
+
+
+def test_train_e2e_fuse_residual(tmp_path, capsys):
+    """Full train() + eval + checkpoint under the deferred-residual
+    block interface (--fuse_residual)."""
+    from vit_10b_fsdp_example_amd.train import main
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    CommContext.reset()
+    cfg = parse_args([
+        "--fake_data", "--fuse_residual", "--image_size", "16",
+        "--patch_size", "4", "--embed_dim", "32", "--num_heads", "2",
+        "--num_blocks", "2", "--num_classes", "10", "--batch_size", "4",
+        "--num_workers", "0", "--num_epochs", "1",
+        "--ckpt_epoch_interval", "1", "--test_epoch_interval", "1",
+        "--log_step_interval", "1", "--warmup_steps", "2",
+        "--max_steps_per_epoch", "3", "--ckpt_dir", str(tmp_path),
+    ])
+    import vit_10b_fsdp_example_amd.data.datasets as ds
+
+    orig_train, orig_val = ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN
+    ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN = 16, 8
+    try:
+        main(cfg)
+    finally:
+        ds.IMAGENET_TRAIN_LEN, ds.IMAGENET_VAL_LEN = orig_train, orig_val
+
+    out = capsys.readouterr().out
+    assert "accuracy on val:" in out and "training completed" in out
+    assert os.path.exists(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))
