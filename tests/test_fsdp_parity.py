@@ -259,3 +259,11 @@ def test_ws2_fsdp_padded_shards():
     for l, g in results:
         np.testing.assert_allclose(l, ref_l, rtol=1e-4, atol=1e-5)
         np.testing.assert_allclose(g, ref_g, rtol=1e-4, atol=1e-5)
+
+
+def test_determinism_same_seed_same_trajectory():
+    """Two identical single-rank runs produce bit-identical loss and
+    grad-norm sequences (seeded init, data and dropout)."""
+    a_l, a_g = _single_rank_reference([], dropout=True)
+    b_l, b_g = _single_rank_reference([], dropout=True)
+    assert a_l == b_l and a_g == b_g

@@ -60,3 +60,15 @@ def test_fallback_to_raw_kernels_table(tmp_path):
     loaded = _load(p)
     assert loaded["k1"] == (2, 3000.0)  # ns summed -> us
     assert loaded["k2"] == (1, 500.0)
+
+
+def test_cli_main_prints_table(tmp_path, capsys):
+    from benchmarks.ktrace_diff import main
+
+    a, b = str(tmp_path / "a.db"), str(tmp_path / "b.db")
+    _make_db(a, [("kern", 4, 400.0, 100.0, 100.0)])
+    _make_db(b, [("kern", 4, 800.0, 200.0, 100.0)])
+    main([a, b, "--top", "5"])
+    out = capsys.readouterr().out
+    assert "B - A: +0.4 ms" in out
+    assert "kern" in out
