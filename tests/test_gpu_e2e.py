@@ -182,3 +182,46 @@ def test_fuse_residual_matches_default_bf16():
     (l0, g0), (l1, g1) = results
     assert abs(l0 - l1) < 1e-3, results
     assert abs(g0 - g1) / max(g0, 1.0) < 2e-2, results
+
+
+def test_shard_on_cpu_training_step():
+    """Host-offload mode on GPU: pinned-host fp32 master shards + CPU
+    AdamW state, device-resident transient full params.  One training
+    step must produce a finite decreasing loss (the 288 GB sizing lever
+    for 60B-class models; ViT-Large offload smoke measured 57 img/s in
+    round 1)."""
+    from vit_10b_fsdp_example_amd.cli import parse_args
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
+    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    CommContext.reset()
+    cfg = parse_args([
+        "--fake_data", "--shard_on_cpu", "--image_size", "32",
+        "--patch_size", "4", "--embed_dim", "256", "--num_heads", "4",
+        "--num_blocks", "2", "--num_classes", "10", "--batch_size", "8",
+        "--num_workers", "0",
+    ])
+    device = xdist.init_distributed()
+    torch.manual_seed(0)
+    model = build_fsdp_vit_model(cfg, device, compute_dtype=torch.bfloat16)
+    # master shards live on (pinned) host memory in this mode
+    shard_devices = {
+        u.flat_param.device.type for u in model._all_units()
+    }
+    assert shard_devices == {"cpu"}, shard_devices
+    loss_fn = CrossEntropyLoss()
+    opt = FusedAdamW(model.parameters(), lr=3e-3, weight_decay=0.1)
+    x = torch.randn(8, 3, 32, 32, device=device, dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (8,), device=device)
+    losses = []
+    for _ in range(6):
+        loss = loss_fn(model(x), y)
+        loss.backward()
+        model.clip_grad_norm_(1.0)
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        losses.append(float(loss))
+    assert all(l == l for l in losses), losses
+    assert losses[-1] < losses[0], losses
