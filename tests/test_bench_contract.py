@@ -51,3 +51,26 @@ def test_bench_default_metric_names_baseline_config():
     src = open(os.path.join(REPO, "bench.py")).read()
     assert "images/sec (whole node) for ViT-10B bs=1024 224px --fake_data" in src
     assert '"vit10b"' in src
+
+
+def test_bench_torchrun_ws2():
+    """The driver's exact N>1 launch shape (torch.distributed.run,
+    one rank per GPU) on CPU/gloo at ws=2: rendezvous, cross-rank MAX
+    timing reduce, exactly one JSON line from rank 0."""
+    port = 29000 + (os.getpid() % 800)
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         "bench.py", "--gpus", "2", "--model", "vit-tiny",
+         "--per_gpu_batch", "2", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, cwd=REPO, timeout=900,
+    )
+    assert res.returncode == 0, res.stderr[-3000:]
+    json_lines = [ln for ln in res.stdout.splitlines()
+                  if ln.startswith("{") and ln.endswith("}")]
+    assert len(json_lines) == 1, res.stdout[-2000:]
+    out = json.loads(json_lines[0])
+    assert out["n_gpus"] == 2
+    assert out["config"]["global_batch"] == 4
+    assert out["config"]["parallelism"] == "fsdp2"
