@@ -129,3 +129,35 @@ def test_train_e2e_fuse_residual(tmp_path, capsys):
     out = capsys.readouterr().out
     assert "accuracy on val:" in out and "training completed" in out
     assert os.path.exists(os.path.join(str(tmp_path), "epoch_1_rank_0.ckpt"))
+
+
+def test_trainer_torchrun_ws2(tmp_path):
+    """run_vit_training.py under real torchrun at ws=2 (gloo): epoch
+    loop + async logger mesh_reduces + per-rank checkpoints + eval,
+    across two actual processes."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    port = 28000 + (os.getpid() % 800)
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         "run_vit_training.py",
+         "--fake_data", "--image_size", "16", "--patch_size", "4",
+         "--embed_dim", "32", "--num_heads", "2", "--num_blocks", "1",
+         "--num_classes", "10", "--batch_size", "4", "--num_workers", "0",
+         "--num_epochs", "1", "--max_steps_per_epoch", "2",
+         "--warmup_steps", "1", "--log_step_interval", "1",
+         "--test_epoch_interval", "1", "--ckpt_dir", str(tmp_path)],
+        capture_output=True, text=True, cwd=repo, timeout=900,
+        env={**os.environ, "VITFSDP_FAKE_LEN": "8"},
+    )
+    assert res.returncode == 0, res.stderr[-3000:]
+    assert "training completed" in res.stdout
+    assert "accuracy on val:" in res.stdout
+    for r in (0, 1):
+        assert os.path.exists(
+            os.path.join(str(tmp_path), f"epoch_1_rank_{r}.ckpt")
+        ), r
