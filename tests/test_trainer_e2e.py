@@ -161,3 +161,28 @@ def test_trainer_torchrun_ws2(tmp_path):
         assert os.path.exists(
             os.path.join(str(tmp_path), f"epoch_1_rank_{r}.ckpt")
         ), r
+
+
+def test_trainer_with_wgrad_dispatch_mode(tmp_path):
+    """VITFSDP_NATIVE_WGRAD=2 in the full trainer: the backward-scoped
+    NativeWgradMode must be harmless end-to-end (on CPU it intercepts
+    nothing — the cuda gate — but the mode still wraps every backward
+    through checkpoint recompute and the FSDP hooks)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = subprocess.run(
+        [sys.executable, "run_vit_training.py",
+         "--fake_data", "--image_size", "16", "--patch_size", "4",
+         "--embed_dim", "32", "--num_heads", "2", "--num_blocks", "1",
+         "--num_classes", "10", "--batch_size", "4", "--num_workers", "0",
+         "--num_epochs", "1", "--max_steps_per_epoch", "2",
+         "--warmup_steps", "1", "--test_epoch_interval", "99",
+         "--ckpt_dir", str(tmp_path)],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+        env={**os.environ, "VITFSDP_FAKE_LEN": "8",
+             "VITFSDP_NATIVE_WGRAD": "2"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    assert "training completed" in res.stdout
