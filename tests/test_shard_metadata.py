@@ -52,3 +52,32 @@ def test_layout_tiles_and_reconstructs(dims, bias):
         torch.testing.assert_close(got, originals[entry["name"]])
     assert covered == info["total_numel"]
     assert set(e["name"] for e in info["params"]) == set(originals)
+
+
+def test_fsdp_rejects_direct_double_wrap():
+    import pytest
+
+    CommContext.reset()
+    inner = FSDP(nn.Linear(4, 4), compute_dtype=torch.float32)
+    with pytest.raises(ValueError, match="do not wrap"):
+        FSDP(inner, compute_dtype=torch.float32)
+
+
+def test_fsdp_rejects_paramless_module():
+    import pytest
+
+    CommContext.reset()
+    with pytest.raises(AssertionError, match="no parameters"):
+        FSDP(nn.ReLU(), compute_dtype=torch.float32)
+
+
+def test_nested_wrap_prunes_inner_units():
+    """A parent FSDP unit must NOT re-shard parameters already owned by
+    a nested FSDP child (the _collect walk prunes them)."""
+    CommContext.reset()
+    child = FSDP(nn.Linear(4, 4), compute_dtype=torch.float32)
+    parent_mod = nn.Sequential(child, nn.Linear(4, 2))
+    parent = FSDP(parent_mod, compute_dtype=torch.float32)
+    # parent's own flat param covers only the outer Linear(4,2): 4*2+2
+    assert parent._total_numel == 10
+    assert child._total_numel == 20
