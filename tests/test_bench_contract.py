@@ -10,6 +10,14 @@ import sys
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 
 def test_bench_json_contract():
     res = subprocess.run(
@@ -57,7 +65,7 @@ def test_bench_torchrun_ws2():
     """The driver's exact N>1 launch shape (torch.distributed.run,
     one rank per GPU) on CPU/gloo at ws=2: rendezvous, cross-rank MAX
     timing reduce, exactly one JSON line from rank 0."""
-    port = 29000 + (os.getpid() % 800)
+    port = _free_port()
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
