@@ -28,7 +28,12 @@ def _worker(rank, world_size, port, fn, args, result_dir):
         raise
 
 
-_PORT = [29600]
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def run_multiprocess(fn, world_size=2, args=(), timeout=300):
@@ -36,8 +41,7 @@ def run_multiprocess(fn, world_size=2, args=(), timeout=300):
 
     Returns the list of per-rank return values (picklable)."""
     ctx = mp.get_context("spawn")
-    _PORT[0] += 1
-    port = _PORT[0] + (os.getpid() % 500)
+    port = _free_port()
     with tempfile.TemporaryDirectory() as result_dir:
         procs = [
             ctx.Process(
