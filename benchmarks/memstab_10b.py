@@ -1,5 +1,10 @@
-import sys, time, torch
-sys.path.insert(0, "/root/repo")
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from vit_10b_fsdp_example_amd import dist as xdist
 from vit_10b_fsdp_example_amd.cli import parse_args
 from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
