@@ -93,3 +93,24 @@ def test_block_grad_ckpt_equivalence():
     )
     for a, b in zip(g1, g2):
         np.testing.assert_allclose(a.numpy(), b.numpy(), rtol=1e-5, atol=1e-6)
+
+
+def test_block_deferred_residual_equivalence():
+    """The deferred-residual pair interface computes exactly the default
+    interface's math across a 2-block chain (the pending add is applied
+    by the NEXT block / the final consumer)."""
+    torch.manual_seed(1)
+    blocks = [Block(dim=32, num_heads=4) for _ in range(2)]
+    deferred = [Block(dim=32, num_heads=4, deferred_residual=True)
+                for _ in range(2)]
+    for d, b in zip(deferred, blocks):
+        d.load_state_dict(b.state_dict())
+
+    x = torch.randn(2, 8, 32)
+    ref = blocks[1](blocks[0](x))
+
+    h, r = deferred[0](x)
+    h, r = deferred[1]((h, r))
+    out = h + r  # final pending add
+    np.testing.assert_allclose(out.detach().numpy(), ref.detach().numpy(),
+                               rtol=1e-5, atol=1e-6)
