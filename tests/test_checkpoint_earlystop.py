@@ -16,18 +16,9 @@ substitute a Function.
 
 import torch
 import torch.nn.functional as F
-from torch.utils._python_dispatch import TorchDispatchMode
+
+from tests.utils_mp import CountMM
 from torch.utils.checkpoint import checkpoint
-
-
-class _CountMM(TorchDispatchMode):
-    def __init__(self):
-        self.n = 0
-
-    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
-        if func._overloadpacket in (torch.ops.aten.mm, torch.ops.aten.addmm):
-            self.n += 1
-        return func(*args, **(kwargs or {}))
 
 
 class _LinearFn(torch.autograd.Function):
@@ -57,7 +48,7 @@ def _backward_gemm_count(use_function):
         return lin(F.gelu(lin(t, w1, b1)), w2, b2)
 
     out = checkpoint(block, x, use_reentrant=False)
-    counter = _CountMM()
+    counter = CountMM()
     with counter:
         out.sum().backward()
     return counter.n

@@ -5,7 +5,8 @@ pins why that matters)."""
 
 import torch
 import torch.nn.functional as F
-from torch.utils._python_dispatch import TorchDispatchMode
+
+from tests.utils_mp import CountMM
 from torch.utils.checkpoint import checkpoint
 
 from vit_10b_fsdp_example_amd.ops import NativeWgradMode
@@ -59,16 +60,6 @@ def test_intercepts_linear_wgrad_and_matches_stock(monkeypatch):
     torch.testing.assert_close(x.grad, ref_x, rtol=1e-6, atol=1e-6)
 
 
-class _CountMM(TorchDispatchMode):
-    def __init__(self):
-        self.n = 0
-
-    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
-        if func._overloadpacket in (torch.ops.aten.mm, torch.ops.aten.addmm):
-            self.n += 1
-        return func(*args, **(kwargs or {}))
-
-
 def test_mode_preserves_checkpoint_early_stop(monkeypatch):
     """The whole point of mode "2": under the dispatch mode the stock
     addmm nodes remain, so the last recompute GEMM is still skipped —
@@ -85,7 +76,7 @@ def test_mode_preserves_checkpoint_early_stop(monkeypatch):
         return F.linear(F.gelu(F.linear(t, w1, b1)), w2, b2)
 
     out = checkpoint(block, x, use_reentrant=False)
-    counter = _CountMM()
+    counter = CountMM()
     wmode = NativeWgradMode(handler=_handler)
     # counter innermost: it sees each ORIGINAL dispatch (then redispatches
     # into wmode), so its count is comparable to the stock-vs-Function

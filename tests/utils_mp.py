@@ -68,3 +68,32 @@ def run_multiprocess(fn, world_size=2, args=(), timeout=300):
                 raise RuntimeError(f"rank {r} failed:\n{payload}")
             results.append(payload)
         return results
+
+
+class CountMM:
+    """TorchDispatchMode counting aten::mm/addmm dispatches (shared by
+    the checkpoint-early-stop and wgrad-dispatch tests)."""
+
+    def __init__(self):
+        self.n = 0
+
+    def __enter__(self):
+        from torch.utils._python_dispatch import TorchDispatchMode
+
+        outer = self
+
+        class _Mode(TorchDispatchMode):
+            def __torch_dispatch__(self, func, types, args=(), kwargs=None):
+                import torch
+
+                if func._overloadpacket in (torch.ops.aten.mm,
+                                            torch.ops.aten.addmm):
+                    outer.n += 1
+                return func(*args, **(kwargs or {}))
+
+        self._mode = _Mode()
+        self._mode.__enter__()
+        return self
+
+    def __exit__(self, *exc):
+        return self._mode.__exit__(*exc)
