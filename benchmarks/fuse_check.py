@@ -33,7 +33,7 @@ for extra in ([], ["--fuse_residual"]):
     loss = CrossEntropyLoss()(m(x), y)
     loss.backward()
     g = m.clip_grad_norm_(1.0)
-    losses.append((float(loss), float(g)))
+    losses.append((float(loss.detach()), float(g)))
 print("default:", losses[0], "fused:", losses[1])
 assert abs(losses[0][0] - losses[1][0]) < 1e-3, "loss mismatch"
 assert abs(losses[0][1] - losses[1][1]) / max(losses[0][1], 1) < 2e-2, "gnorm mismatch"

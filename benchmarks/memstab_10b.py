@@ -27,5 +27,5 @@ for step in range(22):
         torch.cuda.synchronize()
         a = torch.cuda.memory_allocated() / 2**30
         r = torch.cuda.memory_reserved() / 2**30
-        print(f"step {step:2d}: loss {float(loss):.3f} alloc {a:.1f} GiB reserved {r:.1f} GiB", flush=True)
+        print(f"step {step:2d}: loss {float(loss.detach()):.3f} alloc {a:.1f} GiB reserved {r:.1f} GiB", flush=True)
 print("STABLE")

@@ -169,7 +169,7 @@ def test_resume_trajectory_equality(tmp_path):
             opt.step()
             sched.step()
             opt.zero_grad(set_to_none=True)
-            losses.append(float(loss))
+            losses.append(float(loss.detach()))
         return losses
 
     # uninterrupted: 4 + 2 steps

@@ -35,7 +35,7 @@ def test_fsdp_bf16_training_step():
         model.clip_grad_norm_(1.0)
         opt.step()
         opt.zero_grad(set_to_none=True)
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert all(l == l for l in losses), f"NaN loss: {losses}"
     # overfitting a fixed batch must drive the loss down
     assert losses[-1] < losses[0] - 0.3, f"loss not decreasing: {losses}"
@@ -178,7 +178,7 @@ def test_fuse_residual_matches_default_bf16():
         loss = CrossEntropyLoss()(model(x), y)
         loss.backward()
         gn = model.clip_grad_norm_(1.0)
-        results.append((float(loss), float(gn)))
+        results.append((float(loss.detach()), float(gn)))
     (l0, g0), (l1, g1) = results
     assert abs(l0 - l1) < 1e-3, results
     assert abs(g0 - g1) / max(g0, 1.0) < 2e-2, results
@@ -222,6 +222,6 @@ def test_shard_on_cpu_training_step():
         model.clip_grad_norm_(1.0)
         opt.step()
         opt.zero_grad(set_to_none=True)
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert all(l == l for l in losses), losses
     assert losses[-1] < losses[0], losses
