@@ -66,6 +66,7 @@ def _alloc_storage(t, nbytes):
 
 class FullyShardedDataParallel(nn.Module):
     WRAPPER_ATTR = "_fsdp_wrapped_module"
+    _flat_warned = False
 
     def __init__(
         self,
@@ -83,7 +84,18 @@ class FullyShardedDataParallel(nn.Module):
         # flat-by-design: a unit's parameters always live in one flat
         # buffer, because one large RCCL message per unit is the xGMI-
         # efficient shape.  The flag therefore has no effect (documented
-        # deviation; consolidation understands our layout either way).
+        # deviation; consolidation understands our layout either way) —
+        # warn once so passing it is never a silent behavior change.
+        if flatten_parameters and not FullyShardedDataParallel._flat_warned:
+            FullyShardedDataParallel._flat_warned = True
+            import warnings
+
+            warnings.warn(
+                "flatten_parameters is accepted for reference-CLI "
+                "compatibility but has no effect: this FSDP engine is "
+                "flat-by-design (one flat buffer per unit)",
+                stacklevel=2,
+            )
         self.reshard_after_forward = reshard_after_forward
         self.flatten_parameters = flatten_parameters
         self.compute_dtype = compute_dtype
