@@ -9,6 +9,12 @@
 // table rides in the kernarg segment by value and a fixed grid strides
 // over each tensor in turn with float4 (16 B) accesses.  Purely
 // memory-bound; one launch per <=32 tensors.
+//
+// Gradients may arrive in bf16 (the engine's comm-dtype reduced shards,
+// fsdp.py _finalize_unit): the kernels upcast at the register boundary
+// and fold the pending 1/world_size prescale plus the deferred clip
+// coefficient into the same read — the bf16 grad shard is the ONLY
+// gradient memory traffic of the whole step.
 
 #ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
@@ -25,7 +31,7 @@ constexpr int kGrid = 2048;  // 256 CUs x 8 blocks (guide G11)
 
 struct AdamWTable {
   float* p[kMaxTensors];
-  const float* g[kMaxTensors];
+  const void* g[kMaxTensors];  // fp32 or bf16 (GRAD_BF16 template flag)
   float* m[kMaxTensors];
   float* v[kMaxTensors];
   unsigned short* mir[kMaxTensors];  // optional bf16 mirror of p
@@ -34,34 +40,48 @@ struct AdamWTable {
 };
 
 struct TensorTable {
-  float* t[kMaxTensors];
+  void* t[kMaxTensors];  // fp32 or bf16 (per-launch template flag)
   long n[kMaxTensors];
   int count;
 };
 
+template <bool GRAD_BF16>
 __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
                                    float beta2, float eps, float wd,
                                    float bias_c1, float bias_c2,
-                                   const float* __restrict__ grad_scale) {
-  // deferred gradient clipping: the clip coefficient (a device scalar
-  // from clip_grad_norm_(defer_scale=True)) scales g on the fly, which
-  // replaces a separate full read+write pass over every gradient shard
-  const float gs = grad_scale ? *grad_scale : 1.f;
+                                   const float* __restrict__ grad_scale,
+                                   float prescale) {
+  // deferred gradient clipping + mean divide: the clip coefficient (a
+  // device scalar from clip_grad_norm_(defer_scale=True)) and the
+  // 1/world_size prescale scale g on the fly, replacing separate full
+  // read+write passes over every gradient shard
+  const float gs = (grad_scale ? *grad_scale : 1.f) * prescale;
   const float decay = 1.f - lr * wd;
   const float step_size = lr / bias_c1;
   const float inv_sqrt_c2 = rsqrtf(bias_c2);
   for (int ti = 0; ti < tab.count; ++ti) {
     float4* p = (float4*)tab.p[ti];
-    const float4* g = (const float4*)tab.g[ti];
     float4* m = (float4*)tab.m[ti];
     float4* v = (float4*)tab.v[ti];
+    const float4* gf = (const float4*)tab.g[ti];
+    const ushort4* gb = (const ushort4*)tab.g[ti];
     const long n4 = tab.n[ti] / 4;
     for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n4;
          i += (long)gridDim.x * kBlock) {
-      float4 pv = p[i], gv = g[i], mv = m[i], vv = v[i];
+      float4 pv = p[i], mv = m[i], vv = v[i];
+      float gj4[4];
+      if (GRAD_BF16) {
+        ushort4 gv = gb[i];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) gj4[j] = bf16_to_f32((&gv.x)[j]);
+      } else {
+        float4 gv = gf[i];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) gj4[j] = (&gv.x)[j];
+      }
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        float pj = (&pv.x)[j], gj = (&gv.x)[j] * gs;
+        float pj = (&pv.x)[j], gj = gj4[j] * gs;
         float mj = beta1 * (&mv.x)[j] + (1.f - beta1) * gj;
         float vj = beta2 * (&vv.x)[j] + (1.f - beta2) * gj * gj;
         pj *= decay;
@@ -88,7 +108,10 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
     const long tail = tab.n[ti] & 3;
     if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
       const long i = tab.n[ti] - tail + threadIdx.x;
-      float pj = tab.p[ti][i], gj = tab.g[ti][i] * gs;
+      float gj = GRAD_BF16 ? bf16_to_f32(((const unsigned short*)tab.g[ti])[i])
+                           : ((const float*)tab.g[ti])[i];
+      gj *= gs;
+      float pj = tab.p[ti][i];
       float mj = beta1 * tab.m[ti][i] + (1.f - beta1) * gj;
       float vj = beta2 * tab.v[ti][i] + (1.f - beta2) * gj * gj;
       pj = pj * decay - step_size * mj / (sqrtf(vj) * inv_sqrt_c2 + eps);
@@ -100,46 +123,87 @@ __global__ void fused_adamw_kernel(AdamWTable tab, float lr, float beta1,
   }
 }
 
+template <bool BF16>
 __global__ void mt_sqnorm_kernel(TensorTable tab, float* __restrict__ out) {
   __shared__ float scratch[kBlock / WAVE_SIZE];
   float acc = 0.f;
   for (int ti = 0; ti < tab.count; ++ti) {
-    const float4* t = (const float4*)tab.t[ti];
-    const long n4 = tab.n[ti] / 4;
-    for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n4;
-         i += (long)gridDim.x * kBlock) {
-      float4 v = t[i];
-      acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
-    }
-    const long tail = tab.n[ti] & 3;
-    if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
-      float v = tab.t[ti][tab.n[ti] - tail + threadIdx.x];
-      acc += v * v;
+    if (BF16) {
+      // 8 bf16 per 16 B vector load
+      const ushort8_t* t = (const ushort8_t*)tab.t[ti];
+      const long n8 = tab.n[ti] / 8;
+      for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n8;
+           i += (long)gridDim.x * kBlock) {
+        ushort8_t u = t[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf16_to_f32(u[j]);
+          acc += v * v;
+        }
+      }
+      const long tail = tab.n[ti] & 7;
+      if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
+        float v = bf16_to_f32(
+            ((const unsigned short*)tab.t[ti])[tab.n[ti] - tail + threadIdx.x]);
+        acc += v * v;
+      }
+    } else {
+      const float4* t = (const float4*)tab.t[ti];
+      const long n4 = tab.n[ti] / 4;
+      for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n4;
+           i += (long)gridDim.x * kBlock) {
+        float4 v = t[i];
+        acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+      }
+      const long tail = tab.n[ti] & 3;
+      if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
+        float v = ((const float*)tab.t[ti])[tab.n[ti] - tail + threadIdx.x];
+        acc += v * v;
+      }
     }
   }
   acc = block_sum<kBlock / WAVE_SIZE>(acc, scratch);
   if (threadIdx.x == 0) atomicAdd(out, acc);
 }
 
+template <bool BF16>
 __global__ void mt_scale_kernel(TensorTable tab,
                                 const float* __restrict__ factor_ptr,
                                 float factor_imm, int use_ptr) {
   const float f = use_ptr ? *factor_ptr : factor_imm;
   for (int ti = 0; ti < tab.count; ++ti) {
-    float4* t = (float4*)tab.t[ti];
-    const long n4 = tab.n[ti] / 4;
-    for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n4;
-         i += (long)gridDim.x * kBlock) {
-      float4 v = t[i];
-      v.x *= f;
-      v.y *= f;
-      v.z *= f;
-      v.w *= f;
-      t[i] = v;
-    }
-    const long tail = tab.n[ti] & 3;
-    if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
-      tab.t[ti][tab.n[ti] - tail + threadIdx.x] *= f;
+    if (BF16) {
+      ushort8_t* t = (ushort8_t*)tab.t[ti];
+      const long n8 = tab.n[ti] / 8;
+      for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n8;
+           i += (long)gridDim.x * kBlock) {
+        ushort8_t u = t[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) u[j] = f32_to_bf16(bf16_to_f32(u[j]) * f);
+        t[i] = u;
+      }
+      const long tail = tab.n[ti] & 7;
+      if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
+        unsigned short* s =
+            (unsigned short*)tab.t[ti] + (tab.n[ti] - tail + threadIdx.x);
+        *s = f32_to_bf16(bf16_to_f32(*s) * f);
+      }
+    } else {
+      float4* t = (float4*)tab.t[ti];
+      const long n4 = tab.n[ti] / 4;
+      for (long i = (long)blockIdx.x * kBlock + threadIdx.x; i < n4;
+           i += (long)gridDim.x * kBlock) {
+        float4 v = t[i];
+        v.x *= f;
+        v.y *= f;
+        v.z *= f;
+        v.w *= f;
+        t[i] = v;
+      }
+      const long tail = tab.n[ti] & 3;
+      if (tail && blockIdx.x == 0 && threadIdx.x < tail) {
+        ((float*)tab.t[ti])[tab.n[ti] - tail + threadIdx.x] *= f;
+      }
     }
   }
 }
@@ -160,7 +224,8 @@ void fused_adamw(std::vector<torch::Tensor> params,
                  double lr,
                  double beta1, double beta2, double eps, double weight_decay,
                  double bias_c1, double bias_c2,
-                 c10::optional<torch::Tensor> grad_scale) {
+                 c10::optional<torch::Tensor> grad_scale,
+                 double grad_prescale) {
   TORCH_CHECK(mirrors.empty() || mirrors.size() == params.size(),
               "mirrors must be empty or match params");
   const float* gs_ptr = nullptr;
@@ -172,17 +237,25 @@ void fused_adamw(std::vector<torch::Tensor> params,
   }
   auto stream = at::cuda::getCurrentCUDAStream();
   const int n = (int)params.size();
+  if (n == 0) return;
+  const auto grad_dtype = grads[0].scalar_type();
+  TORCH_CHECK(grad_dtype == torch::kFloat32 || grad_dtype == torch::kBFloat16,
+              "fused_adamw: grads must be fp32 or bf16");
   for (int base = 0; base < n; base += kMaxTensors) {
     AdamWTable tab;
     tab.count = std::min(kMaxTensors, n - base);
     long total = 0;
     for (int i = 0; i < tab.count; ++i) {
       auto& p = params[base + i];
+      auto& g = grads[base + i];
       TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32 &&
                       p.is_contiguous(),
                   "fused_adamw expects contiguous fp32 CUDA tensors");
+      TORCH_CHECK(g.scalar_type() == grad_dtype && g.is_contiguous() &&
+                      g.numel() >= p.numel(),
+                  "fused_adamw: grads must share one dtype and cover params");
       tab.p[i] = p.data_ptr<float>();
-      tab.g[i] = grads[base + i].data_ptr<float>();
+      tab.g[i] = g.data_ptr();
       tab.m[i] = exp_avgs[base + i].data_ptr<float>();
       tab.v[i] = exp_avg_sqs[base + i].data_ptr<float>();
       tab.mir[i] = nullptr;
@@ -196,60 +269,95 @@ void fused_adamw(std::vector<torch::Tensor> params,
       tab.n[i] = p.numel();
       total += tab.n[i];
     }
-    hipLaunchKernelGGL(fused_adamw_kernel, dim3(grid_for(total)), dim3(kBlock),
-                       0, stream, tab, (float)lr, (float)beta1, (float)beta2,
-                       (float)eps, (float)weight_decay, (float)bias_c1,
-                       (float)bias_c2, gs_ptr);
+    if (grad_dtype == torch::kBFloat16) {
+      hipLaunchKernelGGL(fused_adamw_kernel<true>, dim3(grid_for(total)),
+                         dim3(kBlock), 0, stream, tab, (float)lr, (float)beta1,
+                         (float)beta2, (float)eps, (float)weight_decay,
+                         (float)bias_c1, (float)bias_c2, gs_ptr,
+                         (float)grad_prescale);
+    } else {
+      hipLaunchKernelGGL(fused_adamw_kernel<false>, dim3(grid_for(total)),
+                         dim3(kBlock), 0, stream, tab, (float)lr, (float)beta1,
+                         (float)beta2, (float)eps, (float)weight_decay,
+                         (float)bias_c1, (float)bias_c2, gs_ptr,
+                         (float)grad_prescale);
+    }
     HIP_CHECK_LAST();
   }
 }
 
+namespace {
+
+// split-by-dtype driver shared by sqnorm and scale: launches the fp32
+// and bf16 template instantiations over homogeneous chunk tables
+template <typename LaunchF32, typename LaunchBF16>
+void mt_dispatch(std::vector<torch::Tensor>& tensors, const char* what,
+                 LaunchF32 launch_f32, LaunchBF16 launch_bf16) {
+  std::vector<torch::Tensor*> f32s, bf16s;
+  for (auto& t : tensors) {
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous(), what,
+                " expects contiguous CUDA tensors");
+    if (t.scalar_type() == torch::kFloat32)
+      f32s.push_back(&t);
+    else if (t.scalar_type() == torch::kBFloat16)
+      bf16s.push_back(&t);
+    else
+      TORCH_CHECK(false, what, ": fp32 or bf16 only");
+  }
+  for (int pass = 0; pass < 2; ++pass) {
+    auto& list = pass == 0 ? f32s : bf16s;
+    const int n = (int)list.size();
+    for (int base = 0; base < n; base += kMaxTensors) {
+      TensorTable tab;
+      tab.count = std::min(kMaxTensors, n - base);
+      long total = 0;
+      for (int i = 0; i < tab.count; ++i) {
+        tab.t[i] = list[base + i]->data_ptr();
+        tab.n[i] = list[base + i]->numel();
+        total += tab.n[i];
+      }
+      if (pass == 0)
+        launch_f32(tab, grid_for(total));
+      else
+        launch_bf16(tab, grid_for(total));
+      HIP_CHECK_LAST();
+    }
+  }
+}
+
+}  // namespace
+
 torch::Tensor multi_tensor_sqnorm(std::vector<torch::Tensor> tensors) {
   auto stream = at::cuda::getCurrentCUDAStream();
   auto out = torch::zeros({}, tensors[0].options().dtype(torch::kFloat32));
-  const int n = (int)tensors.size();
-  for (int base = 0; base < n; base += kMaxTensors) {
-    TensorTable tab;
-    tab.count = std::min(kMaxTensors, n - base);
-    long total = 0;
-    for (int i = 0; i < tab.count; ++i) {
-      auto& t = tensors[base + i];
-      TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
-                      t.is_contiguous(),
-                  "multi_tensor_sqnorm expects contiguous fp32 CUDA tensors");
-      tab.t[i] = t.data_ptr<float>();
-      tab.n[i] = t.numel();
-      total += tab.n[i];
-    }
-    hipLaunchKernelGGL(mt_sqnorm_kernel, dim3(grid_for(total)), dim3(kBlock), 0,
-                       stream, tab, out.data_ptr<float>());
-    HIP_CHECK_LAST();
-  }
+  float* out_ptr = out.data_ptr<float>();
+  mt_dispatch(
+      tensors, "multi_tensor_sqnorm",
+      [&](const TensorTable& tab, int grid) {
+        hipLaunchKernelGGL(mt_sqnorm_kernel<false>, dim3(grid), dim3(kBlock),
+                           0, stream, tab, out_ptr);
+      },
+      [&](const TensorTable& tab, int grid) {
+        hipLaunchKernelGGL(mt_sqnorm_kernel<true>, dim3(grid), dim3(kBlock),
+                           0, stream, tab, out_ptr);
+      });
   return out;
 }
 
 static void mt_scale_impl(std::vector<torch::Tensor>& tensors,
                           const float* factor_ptr, float factor_imm) {
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int n = (int)tensors.size();
-  for (int base = 0; base < n; base += kMaxTensors) {
-    TensorTable tab;
-    tab.count = std::min(kMaxTensors, n - base);
-    long total = 0;
-    for (int i = 0; i < tab.count; ++i) {
-      auto& t = tensors[base + i];
-      TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
-                      t.is_contiguous(),
-                  "multi_tensor_scale expects contiguous fp32 CUDA tensors");
-      tab.t[i] = t.data_ptr<float>();
-      tab.n[i] = t.numel();
-      total += tab.n[i];
-    }
-    hipLaunchKernelGGL(mt_scale_kernel, dim3(grid_for(total)), dim3(kBlock), 0,
-                       stream, tab, factor_ptr, factor_imm,
-                       factor_ptr != nullptr ? 1 : 0);
-    HIP_CHECK_LAST();
-  }
+  const int use_ptr = factor_ptr != nullptr ? 1 : 0;
+  mt_dispatch(
+      tensors, "multi_tensor_scale",
+      [&](const TensorTable& tab, int grid) {
+        hipLaunchKernelGGL(mt_scale_kernel<false>, dim3(grid), dim3(kBlock),
+                           0, stream, tab, factor_ptr, factor_imm, use_ptr);
+      },
+      [&](const TensorTable& tab, int grid) {
+        hipLaunchKernelGGL(mt_scale_kernel<true>, dim3(grid), dim3(kBlock),
+                           0, stream, tab, factor_ptr, factor_imm, use_ptr);
+      });
 }
 
 void multi_tensor_scale(std::vector<torch::Tensor> tensors, double factor) {

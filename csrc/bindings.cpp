@@ -29,7 +29,8 @@ void fused_adamw(std::vector<torch::Tensor> params,
                  double lr,
                  double beta1, double beta2, double eps, double weight_decay,
                  double bias_c1, double bias_c2,
-                 c10::optional<torch::Tensor> grad_scale);
+                 c10::optional<torch::Tensor> grad_scale,
+                 double grad_prescale);
 torch::Tensor multi_tensor_sqnorm(std::vector<torch::Tensor> tensors);
 void multi_tensor_scale(std::vector<torch::Tensor> tensors, double factor);
 void multi_tensor_scale_tensor(std::vector<torch::Tensor> tensors,
@@ -66,7 +67,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + LayerNorm forward");
   m.def("layernorm_add_bwd", &layernorm_add_bwd,
         "fused residual-add + LayerNorm backward");
-  m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (fp32)");
+  m.def("fused_adamw", &fused_adamw,
+        "multi-tensor AdamW step (fp32 master, fp32/bf16 grads with "
+        "folded prescale + deferred clip)",
+        py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
+        py::arg("exp_avg_sqs"), py::arg("mirrors"), py::arg("lr"),
+        py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("weight_decay"), py::arg("bias_c1"), py::arg("bias_c2"),
+        py::arg("grad_scale") = py::none(), py::arg("grad_prescale") = 1.0);
   m.def("multi_tensor_sqnorm", &multi_tensor_sqnorm,
         "sum of squares over tensor list");
   m.def("multi_tensor_scale", &multi_tensor_scale, "in-place scalar scale");

@@ -53,11 +53,11 @@ def test_bench_json_contract():
 
 
 def test_bench_default_metric_names_baseline_config():
-    """With no flags the metric string must name the BASELINE.json
-    headline (ViT-10B bs=1024 224px) — checked statically so we don't
-    build a 10B model on CPU."""
+    """The metric string names the BASELINE.json headline shape with the
+    ACTUAL global batch of the run (bs=1024 at the 8-GPU node point) —
+    checked statically so we don't build a 10B model on CPU."""
     src = open(os.path.join(REPO, "bench.py")).read()
-    assert "images/sec (whole node) for ViT-10B bs=1024 224px --fake_data" in src
+    assert 'f"images/sec (whole node) for ViT-10B bs={b * world} 224px"' in src
     assert '"vit10b"' in src
 
 

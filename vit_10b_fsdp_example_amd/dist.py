@@ -28,6 +28,10 @@ _STATE = {
     "gloo_group": None,  # host-side scalar reduces / barriers
 }
 
+# Collective timeout for the logger side-channel (seconds): bounds how
+# long a rank blocks in mesh_reduce when a peer's closure died.
+_LOGGER_TIMEOUT_S = int(os.environ.get("VITFSDP_LOGGER_TIMEOUT_S", "300"))
+
 
 def init_distributed(device_index=None, timeout_minutes=30):
     """Initialise the per-process distributed runtime.
@@ -41,9 +45,18 @@ def init_distributed(device_index=None, timeout_minutes=30):
     if _STATE["initialized"]:
         return get_device()
 
-    rank = int(os.environ.get("RANK", "0"))
-    world_size = int(os.environ.get("WORLD_SIZE", "1"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if dist.is_available() and dist.is_initialized():
+        # An external harness/launcher already created the process group:
+        # topology truth is the group, not the env.  The gloo
+        # side-channel is still created below so the background logger
+        # never shares a communicator with the training hot path.
+        rank = dist.get_rank()
+        world_size = dist.get_world_size()
+        local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    else:
+        rank = int(os.environ.get("RANK", "0"))
+        world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if device_index is not None:
         local_rank = device_index
 
@@ -70,13 +83,20 @@ def init_distributed(device_index=None, timeout_minutes=30):
             world_size=world_size,
             timeout=datetime.timedelta(minutes=timeout_minutes),
         )
-        if backend == "nccl":
+    if dist.is_initialized():
+        if dist.get_backend() == "gloo":
+            _STATE["gloo_group"] = dist.group.WORLD
+        else:
             # Host-side scalar reduces (loss logging, eval counters) go over
             # gloo so the async logger thread never touches the RCCL
-            # communicator used by the training step.
-            _STATE["gloo_group"] = dist.new_group(backend="gloo")
-        else:
-            _STATE["gloo_group"] = dist.group.WORLD
+            # communicator used by the training step.  Bounded timeout: a
+            # closure that fails on one rank leaves its peers' gather
+            # erroring out after _LOGGER_TIMEOUT_S instead of deadlocking
+            # drain_step_closures() at the epoch boundary.
+            _STATE["gloo_group"] = dist.new_group(
+                backend="gloo",
+                timeout=datetime.timedelta(seconds=_LOGGER_TIMEOUT_S),
+            )
 
     _STATE.update(
         initialized=True,
@@ -218,6 +238,7 @@ class AsyncStepLogger:
         self._queue = queue.Queue()
         self._thread = None
         self._closed = False
+        self.failures = 0
 
     def _ensure_thread(self):
         if self._thread is None:
@@ -235,7 +256,16 @@ class AsyncStepLogger:
                     event.synchronize()
                 fn(*args, **kwargs)
             except Exception as exc:  # pragma: no cover - log, don't kill training
-                print(f"[async-logger] closure failed: {exc!r}", flush=True)
+                # A failed closure on one rank can leave peers blocked in
+                # their logging gather; the gloo side-channel's bounded
+                # timeout (_LOGGER_TIMEOUT_S) turns that into an error on
+                # their logger threads rather than a drain() deadlock.
+                self.failures += 1
+                print(
+                    f"[async-logger] closure failed ({self.failures} total): "
+                    f"{exc!r}",
+                    flush=True,
+                )
             finally:
                 self._queue.task_done()
 

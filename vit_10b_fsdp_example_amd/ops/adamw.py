@@ -7,6 +7,18 @@ fp32 master shards runs as one chunked HIP launch (multi-tensor: a
 chunk table of pointers, decoupled weight decay, fp32 math) instead of
 a python loop of elementwise ops.  CPU path uses torch._foreach with
 identical math, which is what the FSDP-vs-DDP CPU parity tests rely on.
+
+Gradient sources, in priority order per param:
+  * ``p.grad`` — standard fp32 gradient (host-offload path, plain use);
+  * ``p._comm_grad`` + ``p._grad_prescale`` — the FSDP engine's
+    comm-dtype reduced shard with the 1/world_size mean divide left
+    pending; the kernel reads the comm grad, upcasts to fp32 and folds
+    prescale (together with any deferred clip coefficient from
+    clip_grad_norm_(defer_scale=True)) into the same fused pass — no
+    separate cast/divide/scale memory sweeps over the gradients.
+Both attributes are consumed (cleared) by step() and by zero_grad(), so
+a skipped step can never leak a stale gradient or clip coefficient into
+the next iteration.
 """
 
 import math
@@ -28,6 +40,17 @@ class FusedAdamW(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
 
+    @staticmethod
+    def _take_grad(p):
+        """(grad_tensor, prescale) for p, consuming _comm_grad; None if
+        the param has no gradient this step."""
+        if p.grad is not None:
+            return p.grad, 1.0
+        g = getattr(p, "_comm_grad", None)
+        if g is not None:
+            return g, float(getattr(p, "_grad_prescale", 1.0))
+        return None
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
@@ -36,10 +59,12 @@ class FusedAdamW(torch.optim.Optimizer):
                 loss = closure()
 
         for group in self.param_groups:
-            params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
+            params, grads, exp_avgs, exp_avg_sqs, prescales = [], [], [], [], []
             for p in group["params"]:
-                if p.grad is None:
+                got = self._take_grad(p)
+                if got is None:
                     continue
+                g, prescale = got
                 state = self.state[p]
                 if len(state) == 0:
                     state["step"] = 0
@@ -47,15 +72,32 @@ class FusedAdamW(torch.optim.Optimizer):
                     state["exp_avg_sq"] = torch.zeros_like(p)
                 state["step"] += 1
                 params.append(p)
-                grads.append(p.grad)
+                grads.append(g)
+                prescales.append(prescale)
                 exp_avgs.append(state["exp_avg"])
                 exp_avg_sqs.append(state["exp_avg_sq"])
             if not params:
                 continue
 
-            # all params in a group share the step count in practice
-            # (they are stepped together every iteration)
+            # the whole group shares one step count and one prescale:
+            # bias corrections are computed once per launch, so a param
+            # that skipped some steps (intermittent None grad) would
+            # silently get the wrong correction — fail loudly instead
+            # (ADVICE r1).
             step_t = self.state[params[0]]["step"]
+            for p in params:
+                if self.state[p]["step"] != step_t:
+                    raise RuntimeError(
+                        "FusedAdamW: params in one group have diverging "
+                        f"step counts ({self.state[p]['step']} vs {step_t}); "
+                        "partial freezing needs per-param bias correction — "
+                        "put intermittently-frozen params in their own group"
+                    )
+            prescale = prescales[0]
+            if any(s != prescale for s in prescales):
+                raise RuntimeError(
+                    "FusedAdamW: mixed gradient prescales in one group"
+                )
             beta1, beta2 = group["betas"]
             bias_c1 = 1.0 - beta1 ** step_t
             bias_c2 = 1.0 - beta2 ** step_t
@@ -91,15 +133,22 @@ class FusedAdamW(torch.optim.Optimizer):
                     bias_c1,
                     bias_c2,
                     grad_scale,
+                    prescale,
                 )
                 for p, m in zip(params, mirrors or []):
                     if m is not None:
                         p._mirror_fresh = True
             else:
-                if grad_scale is not None:
-                    grads = torch._foreach_mul(grads, float(grad_scale))
+                eff = grads
+                if grad_scale is not None or prescale != 1.0:
+                    s = prescale * (
+                        float(grad_scale) if grad_scale is not None else 1.0
+                    )
+                    eff = [g.to(torch.float32) * s for g in grads]
+                elif any(g.dtype != torch.float32 for g in grads):
+                    eff = [g.to(torch.float32) for g in grads]
                 self._foreach_step(
-                    params, grads, exp_avgs, exp_avg_sqs,
+                    params, eff, exp_avgs, exp_avg_sqs,
                     group["lr"], beta1, beta2, group["eps"],
                     group["weight_decay"], bias_c1, bias_c2,
                 )
@@ -108,7 +157,25 @@ class FusedAdamW(torch.optim.Optimizer):
                     if m is not None:
                         m.copy_(p.detach().to(m.dtype))
                         p._mirror_fresh = True
+
+            # comm grads are single-use: drop the references so the
+            # buffers free (ws=1 hands the engine's reusable flat
+            # buffer, ws>1 per-unit shards)
+            for p in params:
+                if getattr(p, "_comm_grad", None) is not None:
+                    p._comm_grad = None
         return loss
+
+    @torch.no_grad()
+    def zero_grad(self, set_to_none=True):
+        super().zero_grad(set_to_none=set_to_none)
+        for group in self.param_groups:
+            for p in group["params"]:
+                # a skipped step must not leak gradient state forward
+                if getattr(p, "_comm_grad", None) is not None:
+                    p._comm_grad = None
+                if getattr(p, "_deferred_grad_scale", None) is not None:
+                    p._deferred_grad_scale = None
 
     @staticmethod
     def _foreach_step(params, grads, exp_avgs, exp_avg_sqs, lr, beta1, beta2,

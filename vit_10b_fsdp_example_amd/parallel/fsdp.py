@@ -188,9 +188,24 @@ class FullyShardedDataParallel(nn.Module):
         self.register_load_state_dict_post_hook(self._invalidate_mirror_hook)
 
         # ---- full compute-dtype buffer + stable leaf views ----
-        self._full_flat = torch.empty(
-            self._padded_numel, dtype=compute_dtype, device=self.device
-        )
+        # ws=1 short-circuit: with one rank the "gathered full params"
+        # and the bf16 comm mirror are the same numbers, so the full
+        # buffer ALIASES the mirror's storage — no gather copy, no
+        # separate 2x-params residency, and the optimizer's mirror
+        # refresh IS the weight update the next forward reads.  The
+        # full buffer is then persistent (never storage-freed).
+        self._ws1_alias = ws == 1 and self._mirror is not None
+        if self._ws1_alias:
+            # alias with its own autograd version counter (mirror
+            # refreshes via copy_ must not bump the param views' base)
+            self._full_flat = torch.empty(0, dtype=compute_dtype, device=self.device)
+            self._full_flat.set_(
+                self._mirror.untyped_storage(), 0, (self._padded_numel,)
+            )
+        else:
+            self._full_flat = torch.empty(
+                self._padded_numel, dtype=compute_dtype, device=self.device
+            )
         # Collective writes go through a separate alias tensor with its
         # own autograd version counter, so refilling the buffer in the
         # pre-backward hook never trips saved-tensor version checks on
@@ -211,7 +226,8 @@ class FullyShardedDataParallel(nn.Module):
                 setattr(mod, name, view)
                 self._param_infos.append((mod, name, shape, numel, off))
                 self._views.append(view)
-        _free_storage(self._full_flat)
+        if not self._ws1_alias:
+            _free_storage(self._full_flat)
 
         for idx, view in enumerate(self._views):
             view.register_post_accumulate_grad_hook(self._make_grad_hook(idx))
@@ -272,6 +288,13 @@ class FullyShardedDataParallel(nn.Module):
 
     def _materialize(self):
         """Ensure full params are resident and up to date."""
+        if self._ws1_alias:
+            # full params ARE the mirror; just make sure it is fresh
+            # (FusedAdamW refreshes it in its step epilogue, so this is
+            # a no-op in the steady state)
+            self._comm_shard()
+            self._fresh = True
+            return
         if self._pending_gather is not None:
             work, _src = self._pending_gather
             self._pending_gather = None
@@ -287,11 +310,15 @@ class FullyShardedDataParallel(nn.Module):
     def _prefetch_gather(self):
         """Issue this unit's param gather asynchronously (called by the
         previous unit in execution order while it computes)."""
+        if self._ws1_alias:
+            return
         if self._pending_gather is not None or (self._resident() and self._fresh):
             return
         self._pending_gather = self._issue_gather(async_op=True)
 
     def _free_full(self):
+        if self._ws1_alias:
+            return  # persistent alias of the mirror
         _free_storage(self._full_flat)
 
     # ------------------------------------------------------------------
@@ -373,6 +400,16 @@ class FullyShardedDataParallel(nn.Module):
                 self._finalize_backward_all
             )
         self._materialize()
+        # ws=1 hands _full_grad itself to the optimizer as _comm_grad; if
+        # a second backward starts before the step consumed it (gradient
+        # accumulation), detach the pending grad before zeroing the buffer
+        pending = getattr(self.flat_param, "_comm_grad", None)
+        if (
+            pending is not None
+            and self._full_grad.untyped_storage().size() > 0
+            and pending.data_ptr() == self._full_grad.data_ptr()
+        ):
+            self.flat_param._comm_grad = pending.clone()
         _alloc_storage(self._full_grad, self._padded_numel * self._elem_bytes)
         self._full_grad.zero_()
         # Pre-bind each view's .grad to its slice of the flat grad buffer:
@@ -416,6 +453,12 @@ class FullyShardedDataParallel(nn.Module):
         full params and launch the async reduce-scatter."""
         self._free_full()
         self._fresh = False  # master shard will change at optimizer.step
+        if self._comm.world_size == 1:
+            # single rank: the "reduced shard" IS the flat grad buffer —
+            # no copy, no extra allocation (the buffer is handed to the
+            # optimizer below and its storage is reused next backward)
+            self._pending_reduce = (None, self._full_grad)
+            return
         out_shard = torch.empty(
             self._shard_numel, dtype=self.compute_dtype, device=self.device
         )
@@ -423,8 +466,17 @@ class FullyShardedDataParallel(nn.Module):
         self._pending_reduce = (work, out_shard)
 
     def _finalize_unit(self):
-        """Wait the pending reduction, install the fp32 mean grad on the
-        master shard, release buffers."""
+        """Wait the pending reduction and hand the reduced grad shard to
+        the optimizer, releasing buffers.
+
+        The reduced shard stays in comm/compute dtype: it is attached as
+        ``flat_param._comm_grad`` with ``flat_param._grad_prescale`` =
+        1/world_size, and FusedAdamW folds the prescale (together with
+        any deferred clip coefficient) into its fp32 gradient read —
+        saving the cast + divide memory passes over every gradient every
+        step.  The host-offload path (--shard_on_cpu) keeps the explicit
+        fp32 D2H ingestion since the master lives on another device.
+        """
         if self._grads_arrived and self._pending_reduce is None:
             # partial grads (frozen subgraph): reduce what we have —
             # missing slices are zeros from the buffer memset.
@@ -432,21 +484,50 @@ class FullyShardedDataParallel(nn.Module):
         if self._pending_reduce is not None:
             work, out_shard = self._pending_reduce
             self._pending_reduce = None
-            work.wait()
-            g = out_shard.to(torch.float32)
-            if self._comm.world_size > 1:
-                g.div_(self._comm.world_size)
-            if g.device != self.flat_param.device:
-                g = g.to(self.flat_param.device)  # host-offload: grad D2H
-            if self.flat_param.grad is None:
-                self.flat_param.grad = g
+            if work is not None:
+                work.wait()
+            prescale = 1.0 / self._comm.world_size
+            fp32_ingest = (
+                out_shard.device != self.flat_param.device  # --shard_on_cpu D2H
+                or os.environ.get("VITFSDP_FP32_GRAD_INGEST", "0") == "1"
+            )
+            if fp32_ingest:
+                # explicit fp32 .grad: required for host offload (the
+                # master lives on another device) and as the escape hatch
+                # for optimizers that only read .grad (anything other
+                # than ops.FusedAdamW).  copy=True because at ws=1
+                # out_shard aliases _full_grad, which is freed below.
+                g = out_shard.to(torch.float32, copy=True)
+                if prescale != 1.0:
+                    g.mul_(prescale)
+                g = g.to(self.flat_param.device)
+                if self.flat_param.grad is None:
+                    self.flat_param.grad = g
+                else:
+                    self.flat_param.grad.add_(g)
+            elif getattr(self.flat_param, "_comm_grad", None) is not None:
+                # gradient accumulation across backwards: fold the new
+                # shard into the existing one in fp32 (rare path)
+                acc = self.flat_param._comm_grad
+                if acc.dtype != torch.float32:
+                    acc = acc.to(torch.float32)
+                acc.add_(out_shard.to(torch.float32))
+                self.flat_param._comm_grad = acc
+                self.flat_param._grad_prescale = prescale
             else:
-                self.flat_param.grad.add_(g)
+                self.flat_param._comm_grad = out_shard
+                self.flat_param._grad_prescale = prescale
             # the optimizer will rewrite the master next; unless it also
             # refreshes the mirror (FusedAdamW does), the mirror is stale
             if self._mirror is not None:
                 self.flat_param._mirror_fresh = False
-        _free_storage(self._full_grad)
+        if (
+            getattr(self.flat_param, "_comm_grad", None) is not None
+            and self.flat_param._comm_grad.data_ptr() == self._full_grad.data_ptr()
+        ):
+            pass  # ws=1: the optimizer consumes the buffer in place
+        else:
+            _free_storage(self._full_grad)
         self._grads_arrived = 0
 
     def _finalize_backward_all(self):
@@ -471,11 +552,28 @@ class FullyShardedDataParallel(nn.Module):
         bytes memory pass per step).  Only use defer_scale with an
         optimizer that honors _deferred_grad_scale (ops.FusedAdamW)."""
         assert norm_type == 2.0, "only L2 clipping is supported"
-        units = [u for u in self._all_units() if u.flat_param.grad is not None]
-        grads = [u.flat_param.grad for u in units]
-        if not grads:
+        # gradients arrive either as fp32 .grad (host-offload path) or as
+        # comm-dtype _comm_grad with a pending 1/ws prescale that
+        # FusedAdamW folds into its read (_finalize_unit)
+        units, grads, raw, raw_prescale = [], [], [], 1.0
+        for u in self._all_units():
+            p = u.flat_param
+            if p.grad is not None:
+                units.append(u)
+                grads.append(p.grad)
+            elif getattr(p, "_comm_grad", None) is not None:
+                units.append(u)
+                raw.append(p._comm_grad)
+                raw_prescale = p._grad_prescale
+        if not units:
             return torch.zeros((), device=self.device)
-        local = local_sqnorm(grads)
+        local = None
+        if grads:
+            local = local_sqnorm(grads)
+        if raw:
+            # ||prescale * g||^2 == prescale^2 * ||g||^2
+            r = local_sqnorm(raw) * (raw_prescale * raw_prescale)
+            local = r if local is None else local + r
         if local.device != self.device:
             local = local.to(self.device)
         self._comm.all_reduce_scalar_(local)
@@ -484,9 +582,19 @@ class FullyShardedDataParallel(nn.Module):
         clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0)
         if defer_scale and self.device.type == "cuda":
             for u in units:
+                if getattr(u.flat_param, "_deferred_grad_scale", None) is not None:
+                    raise RuntimeError(
+                        "clip_grad_norm_(defer_scale=True): the previous "
+                        "deferred clip coefficient was never consumed — "
+                        "defer_scale requires an optimizer that honors "
+                        "_deferred_grad_scale every step (ops.FusedAdamW)"
+                    )
                 u.flat_param._deferred_grad_scale = clip_coef
         else:
-            scale_(grads, clip_coef)
+            if grads:
+                scale_(grads, clip_coef)
+            if raw:
+                scale_(raw, clip_coef)
         return total_norm
 
     # ------------------------------------------------------------------
