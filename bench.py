@@ -48,7 +48,7 @@ def main():
     from vit_10b_fsdp_example_amd.cli import parse_args as cli_parse
     from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
     from vit_10b_fsdp_example_amd.ops import (
-        CrossEntropyLoss, FusedAdamW, wgrad_backward_context,
+        CrossEntropyLoss, FusedAdamW, gemm_dispatch_context,
     )
     from vit_10b_fsdp_example_amd.utils import get_warmup_cosine_scheduler
 
@@ -102,8 +102,10 @@ def main():
 
     def one_step(i):
         x, y = batches[i % 2]
-        loss = loss_fn(model(x), y)
-        with wgrad_backward_context():
+        # dispatch-level GEMM rerouting (tuned hipBLASLt indices /
+        # native wgrad) covers forward, recompute and backward
+        with gemm_dispatch_context():
+            loss = loss_fn(model(x), y)
             loss.backward()
         if cfg.clip_grad_norm > 0:
             model.clip_grad_norm_(cfg.clip_grad_norm, defer_scale=True)

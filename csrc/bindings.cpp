@@ -58,6 +58,13 @@ std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
 torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
                       c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> lt_gemm_gelu(torch::Tensor a, torch::Tensor b,
+                                        c10::optional<torch::Tensor> bias,
+                                        long algo_index);
+std::vector<torch::Tensor> lt_gemm_dgelu_bgrad(torch::Tensor dy,
+                                               torch::Tensor w,
+                                               torch::Tensor aux,
+                                               long algo_index);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native CDNA4 kernels for the FSDP ViT framework";
@@ -100,4 +107,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "csrc/tools/hipblaslt_search.cpp",
         py::arg("a"), py::arg("b"), py::arg("algo_index") = -1,
         py::arg("bias") = py::none());
+  m.def("lt_gemm_gelu", &lt_gemm_gelu,
+        "(gelu(a@b+bias), pre-activation aux) in one hipblaslt-ext GEMM "
+        "(GELU_AUX epilogue; tanh-approx GELU)",
+        py::arg("a"), py::arg("b"), py::arg("bias") = py::none(),
+        py::arg("algo_index") = -1);
+  m.def("lt_gemm_dgelu_bgrad", &lt_gemm_dgelu_bgrad,
+        "(dgelu(dy@w, aux), column-sum bias grad) in one hipblaslt-ext "
+        "GEMM (DGELU_BGRAD epilogue)",
+        py::arg("dy"), py::arg("w"), py::arg("aux"),
+        py::arg("algo_index") = -1);
 }

@@ -1,19 +1,27 @@
-// hipBLASLt invocation with an explicit algorithm index (ROADMAP item 5).
+// hipBLASLt invocations with explicit algorithm indices and fused
+// epilogues (ROADMAP items 5/6).
 //
 // PyTorch's TunableOp crashes this ROCm 7.2 stack, so algorithm
-// selection is done offline by csrc/tools/hipblaslt_search.cpp; this
-// host-side entry applies a chosen index from Python:
+// selection is done offline by csrc/tools/hipblaslt_search.cpp; the
+// entries here apply a chosen index from Python:
 //
-//   d = _C.lt_gemm(a, b, algo_index)   # row-major a[M,K] @ b[K,N]
+//   d        = _C.lt_gemm(a, b, algo_index, bias)        # d = a @ b (+bias)
+//   d, aux   = _C.lt_gemm_gelu(a, b, bias, algo_index)   # d = gelu(a@b+bias)
+//   dx, dbia = _C.lt_gemm_dgelu_bgrad(dy, w, aux, algo_index)
 //
 // algo_index < 0 uses the library heuristic (same pick as torch's
-// matmul) — giving a GPU test that validates the whole hipblaslt-ext
-// path (and the row/column-major duality) against torch.matmul before
-// any tuned index is wired in.
+// matmul).  a/b may be row-contiguous tensors OR transposed views of
+// row-contiguous tensors — both map onto hipBLASLt op flags with zero
+// copies, which is what lets the dispatch-mode router (ops/linear.py)
+// reroute the training GEMMs (forward x@W^T, dgrad dy@W, wgrad dy^T@x)
+// without materializing any transpose.
 //
-// Row-major C = A @ B is computed as the column-major dual
-// C^T = B^T A^T: opA = opB = N, (m, n, k) = (N, M, K), A-ptr = b,
-// B-ptr = a, leading dims (N, K, N).
+// Layout math: row-major C[M,N] = A @ B is computed as the column-major
+// dual D[N,M] = opA(B_mem) @ opB(A_mem): a row-contiguous operand is
+// its own transpose in column-major interpretation (op=N), a transposed
+// view needs op=T; leading dims are the underlying row lengths.  This
+// reproduces exactly the Tensile kernel families the trace shows
+// (Cijk_Alik TN fwd, Cijk_Ailk NN dgrad, Cijk_Ailk_Bjlk NT wgrad).
 
 #ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
@@ -54,59 +62,43 @@ void* lt_workspace(size_t bytes) {
 
 constexpr size_t kMaxWorkspace = 128u << 20;
 
-}  // namespace
+// 2-D operand layout: row-contiguous (op N in the column-major dual) or
+// a transposed view of a row-contiguous base (op T).  Returns false if
+// neither (caller must materialize).
+struct Operand {
+  hipblasOperation_t op;
+  int64_t ld;
+  const void* ptr;
+};
 
-torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
-                      c10::optional<torch::Tensor> bias) {
-  TORCH_CHECK(a.is_cuda() && b.is_cuda(), "lt_gemm: CUDA tensors required");
-  TORCH_CHECK(a.dtype() == torch::kBFloat16 && b.dtype() == torch::kBFloat16,
-              "lt_gemm: bf16 only");
-  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0),
-              "lt_gemm: a[M,K] @ b[K,N] expected");
-  auto ac = a.contiguous();
-  auto bc = b.contiguous();
-  const int64_t M = ac.size(0), K = ac.size(1), N = bc.size(1);
-  auto d = torch::empty({M, N}, ac.options());
-
-  auto handle = lt_handle();
-  hipStream_t stream = at::cuda::getCurrentCUDAStream();
-
-  hipblaslt_ext::Gemm gemm(handle, HIPBLAS_OP_N, HIPBLAS_OP_N, HIP_R_16BF,
-                           HIP_R_16BF, HIP_R_16BF, HIP_R_16BF,
-                           HIPBLAS_COMPUTE_32F);
-  float alpha = 1.0f, beta = 0.0f;
-  hipblaslt_ext::GemmEpilogue epilogue;
-  hipblaslt_ext::GemmInputs inputs;
-  torch::Tensor bias_c;
-  if (bias.has_value()) {
-    // per-output-feature bias: length N = rows of the column-major dual,
-    // which is exactly hipBLASLt's bias-vector broadcast
-    bias_c = bias->contiguous();
-    TORCH_CHECK(bias_c.is_cuda() && bias_c.dtype() == torch::kBFloat16 &&
-                    bias_c.numel() == N,
-                "lt_gemm: bias must be bf16 CUDA of length N");
-    epilogue.setMode(HIPBLASLT_EPILOGUE_BIAS);
-    epilogue.setBiasDataType(HIP_R_16BF);
-    inputs.setBias(bias_c.data_ptr());
+bool classify(const torch::Tensor& t, Operand& out) {
+  const int64_t r = t.size(0), c = t.size(1);
+  const int64_t s0 = t.stride(0), s1 = t.stride(1);
+  out.ptr = t.data_ptr();
+  // row-contiguous [r, c]: column-major read gives the transpose we
+  // need directly
+  if (s1 == 1 && (s0 == c || r == 1)) {
+    out.op = HIPBLAS_OP_N;
+    out.ld = c;  // underlying row length
+    return true;
   }
-  inputs.setA(bc.data_ptr());  // column-major dual: A <- b
-  inputs.setB(ac.data_ptr());
-  inputs.setC(d.data_ptr());
-  inputs.setD(d.data_ptr());
-  inputs.setAlpha(&alpha);
-  inputs.setBeta(&beta);
-  // (m, n, k) = (N, M, K); short setProblem overload derives the
-  // contiguous leading dims (N, K, N)
-  TORCH_CHECK(
-      gemm.setProblem(N, M, K, 1, epilogue, inputs) == HIPBLAS_STATUS_SUCCESS,
-      "lt_gemm: setProblem failed");
+  // transposed view of a row-contiguous [c, r] base
+  if (s0 == 1 && (s1 == r || c == 1)) {
+    out.op = HIPBLAS_OP_T;
+    out.ld = r;
+    return true;
+  }
+  return false;
+}
 
+hipblasLtMatmulAlgo_t pick_algo(hipblaslt_ext::Gemm& gemm, long algo_index,
+                                const char* what) {
   hipblasLtMatmulAlgo_t algo;
-  bool have_algo = false;
   if (algo_index >= 0) {
     std::vector<hipblasLtMatmulHeuristicResult_t> found;
     std::vector<int> idx{static_cast<int>(algo_index)};
-    if (hipblaslt_ext::getAlgosFromIndex(handle, idx, found) ==
+    bool ok = false;
+    if (hipblaslt_ext::getAlgosFromIndex(lt_handle(), idx, found) ==
             HIPBLAS_STATUS_SUCCESS &&
         !found.empty()) {
       size_t need = 0;
@@ -114,29 +106,153 @@ torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
               HIPBLAS_STATUS_SUCCESS &&
           need <= kMaxWorkspace) {
         algo = found[0].algo;
-        have_algo = true;
+        ok = true;
       }
     }
-    TORCH_CHECK(have_algo, "lt_gemm: algo index ", algo_index,
+    TORCH_CHECK(ok, what, ": algo index ", algo_index,
                 " not valid for this problem");
-  } else {
-    hipblaslt_ext::GemmPreference pref;
-    pref.setMaxWorkspaceBytes(kMaxWorkspace);
-    std::vector<hipblasLtMatmulHeuristicResult_t> heur;
-    TORCH_CHECK(gemm.algoGetHeuristic(1, pref, heur) ==
-                        HIPBLAS_STATUS_SUCCESS &&
-                    !heur.empty(),
-                "lt_gemm: no heuristic algorithm found");
-    algo = heur[0].algo;
-    have_algo = true;
+    return algo;
   }
+  hipblaslt_ext::GemmPreference pref;
+  pref.setMaxWorkspaceBytes(kMaxWorkspace);
+  std::vector<hipblasLtMatmulHeuristicResult_t> heur;
+  TORCH_CHECK(
+      gemm.algoGetHeuristic(1, pref, heur) == HIPBLAS_STATUS_SUCCESS &&
+          !heur.empty(),
+      what, ": no heuristic algorithm found");
+  return heur[0].algo;
+}
 
+// Shared driver: row-major d[M,N] = a[M,K] @ b[K,N] with an optional
+// fused epilogue.  bias: input vector length N (BIAS-family modes) or
+// OUTPUT vector length N (BGRAD-family modes).  aux: [M,N] bf16, input
+// for DGELU*, output for *_AUX_* modes.
+torch::Tensor lt_run(const torch::Tensor& a, const torch::Tensor& b,
+                     long algo_index, hipblasLtEpilogue_t mode,
+                     void* bias_ptr, void* aux_ptr, const char* what) {
+  TORCH_CHECK(a.is_cuda() && b.is_cuda(), what, ": CUDA tensors required");
+  TORCH_CHECK(a.dtype() == torch::kBFloat16 && b.dtype() == torch::kBFloat16,
+              what, ": bf16 only");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0), what,
+              ": a[M,K] @ b[K,N] expected");
+  const int64_t M = a.size(0), K = a.size(1), N = b.size(1);
+
+  Operand opa, opb;
+  torch::Tensor a_hold = a, b_hold = b;
+  if (!classify(a_hold, opa)) {
+    a_hold = a.contiguous();
+    classify(a_hold, opa);
+  }
+  if (!classify(b_hold, opb)) {
+    b_hold = b.contiguous();
+    classify(b_hold, opb);
+  }
+  auto d = torch::empty({M, N}, a_hold.options());
+
+  auto handle = lt_handle();
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+
+  // column-major dual: A-slot <- b memory, B-slot <- a memory
+  hipblaslt_ext::Gemm gemm(handle, opb.op, opa.op, HIP_R_16BF, HIP_R_16BF,
+                           HIP_R_16BF, HIP_R_16BF, HIPBLAS_COMPUTE_32F);
+  float alpha = 1.0f, beta = 0.0f;
+  hipblaslt_ext::GemmEpilogue epilogue;
+  hipblaslt_ext::GemmInputs inputs;
+  if (mode != HIPBLASLT_EPILOGUE_DEFAULT) {
+    epilogue.setMode(mode);
+    if (bias_ptr != nullptr) epilogue.setBiasDataType(HIP_R_16BF);
+    if (aux_ptr != nullptr) {
+      epilogue.setAuxDataType(HIP_R_16BF);
+      epilogue.setAuxLeadingDimension((int)N);
+      epilogue.setAuxBatchStride((int)(M * N));
+    }
+  }
+  if (bias_ptr != nullptr) inputs.setBias(bias_ptr);
+  if (aux_ptr != nullptr) inputs.setAux(aux_ptr);
+  inputs.setA(const_cast<void*>(opb.ptr));
+  inputs.setB(const_cast<void*>(opa.ptr));
+  inputs.setC(d.data_ptr());
+  inputs.setD(d.data_ptr());
+  inputs.setAlpha(&alpha);
+  inputs.setBeta(&beta);
+
+  hipblaslt_ext::GemmProblemType ptype(opb.op, opa.op, HIP_R_16BF, HIP_R_16BF,
+                                       HIP_R_16BF, HIP_R_16BF,
+                                       HIPBLAS_COMPUTE_32F);
+  // (m, n, k) = (N, M, K); lda/ldb are the operands' underlying row
+  // lengths, ldc/ldd = N
+  TORCH_CHECK(gemm.setProblem(N, M, K, 1, opb.ld, opa.ld, N, N, 0, 0, 0, 0,
+                              epilogue, inputs,
+                              ptype) == HIPBLAS_STATUS_SUCCESS,
+              what, ": setProblem failed");
+
+  auto algo = pick_algo(gemm, algo_index, what);
   TORCH_CHECK(gemm.initialize(algo, lt_workspace(kMaxWorkspace), false,
                               stream) == HIPBLAS_STATUS_SUCCESS,
-              "lt_gemm: initialize failed");
-  TORCH_CHECK(gemm.run(stream) == HIPBLAS_STATUS_SUCCESS,
-              "lt_gemm: run failed");
+              what, ": initialize failed");
+  TORCH_CHECK(gemm.run(stream) == HIPBLAS_STATUS_SUCCESS, what,
+              ": run failed");
   return d;
+}
+
+}  // namespace
+
+torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
+                      c10::optional<torch::Tensor> bias) {
+  void* bias_ptr = nullptr;
+  torch::Tensor bias_c;
+  hipblasLtEpilogue_t mode = HIPBLASLT_EPILOGUE_DEFAULT;
+  if (bias.has_value()) {
+    bias_c = bias->contiguous();
+    TORCH_CHECK(bias_c.is_cuda() && bias_c.dtype() == torch::kBFloat16 &&
+                    bias_c.numel() == b.size(1),
+                "lt_gemm: bias must be bf16 CUDA of length N");
+    bias_ptr = bias_c.data_ptr();
+    mode = HIPBLASLT_EPILOGUE_BIAS;
+  }
+  return lt_run(a, b, algo_index, mode, bias_ptr, nullptr, "lt_gemm");
+}
+
+// d = gelu(a @ b + bias), aux = a @ b + bias (the pre-activation, saved
+// for the backward's DGELU epilogue).  NOTE: hipBLASLt's GELU is the
+// tanh approximation; numerics policy handled by the Python caller.
+std::vector<torch::Tensor> lt_gemm_gelu(torch::Tensor a, torch::Tensor b,
+                                        c10::optional<torch::Tensor> bias,
+                                        long algo_index) {
+  const int64_t M = a.size(0), N = b.size(1);
+  auto aux = torch::empty({M, N}, a.options().dtype(torch::kBFloat16));
+  void* bias_ptr = nullptr;
+  torch::Tensor bias_c;
+  hipblasLtEpilogue_t mode = HIPBLASLT_EPILOGUE_GELU_AUX;
+  if (bias.has_value()) {
+    bias_c = bias->contiguous();
+    TORCH_CHECK(bias_c.is_cuda() && bias_c.dtype() == torch::kBFloat16 &&
+                    bias_c.numel() == N,
+                "lt_gemm_gelu: bias must be bf16 CUDA of length N");
+    bias_ptr = bias_c.data_ptr();
+    mode = HIPBLASLT_EPILOGUE_GELU_AUX_BIAS;
+  }
+  auto d = lt_run(a, b, algo_index, mode, bias_ptr, aux.data_ptr(),
+                  "lt_gemm_gelu");
+  return {d, aux};
+}
+
+// dx = dgelu(dy @ w, aux) with dbias = column-sums of dx fused
+// (DGELU_BGRAD): the backward of h = gelu(pre) where pre = x@W1^T+b1 —
+// dy@w is d(gelu_out), aux is pre, dx is d(pre), dbias is db1.
+std::vector<torch::Tensor> lt_gemm_dgelu_bgrad(torch::Tensor dy,
+                                               torch::Tensor w,
+                                               torch::Tensor aux,
+                                               long algo_index) {
+  const int64_t M = dy.size(0), N = w.size(1);
+  TORCH_CHECK(aux.is_cuda() && aux.dtype() == torch::kBFloat16 &&
+                  aux.dim() == 2 && aux.size(0) == M && aux.size(1) == N &&
+                  aux.is_contiguous(),
+              "lt_gemm_dgelu_bgrad: aux must be contiguous bf16 [M,N]");
+  auto dbias = torch::empty({N}, dy.options().dtype(torch::kBFloat16));
+  auto d = lt_run(dy, w, algo_index, HIPBLASLT_EPILOGUE_DGELU_BGRAD,
+                  dbias.data_ptr(), aux.data_ptr(), "lt_gemm_dgelu_bgrad");
+  return {d, dbias};
 }
 
 #endif  // VITFSDP_KERNELS_ONLY

@@ -57,9 +57,10 @@ struct Problem {
     int64_t m, n, k;
 };
 
-// ViT-10B, bs=128/GPU, 224px: tokens = 128 * 257 = 32896.
+// ViT-10B, bs=128/GPU, 224px: tokens = 128 * 256 = 32768 (the model is
+// mean-pooled, no CLS token — T = (224/14)^2, models/vit.py).
 static std::vector<Problem> problems_vit10b() {
-    const int64_t tok = 32896, d = 5120, qkv = 15360, ffn = 20480;
+    const int64_t tok = 32768, d = 5120, qkv = 15360, ffn = 20480;
     return {
         {"patch_fwd", HIPBLAS_OP_T, HIPBLAS_OP_N, d, tok, 588},
         {"qkv_fwd", HIPBLAS_OP_T, HIPBLAS_OP_N, qkv, tok, d},

@@ -1,0 +1,140 @@
+"""Contract tests for the comm-dtype gradient handoff between the FSDP
+engine and FusedAdamW (fsdp.py _finalize_unit <-> ops/adamw.py):
+
+  * after backward the reduced shard sits in flat_param._comm_grad (with
+    _grad_prescale = 1/ws) and flat_param.grad stays None;
+  * FusedAdamW.step consumes it (identical math to explicit fp32
+    ingestion) and clears the reference;
+  * zero_grad() discards a pending comm grad so a skipped step cannot
+    leak gradient state forward;
+  * VITFSDP_FP32_GRAD_INGEST=1 restores .grad for foreign optimizers;
+  * per-group step-count divergence in FusedAdamW raises loudly.
+"""
+
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+
+from vit_10b_fsdp_example_amd.ops import FusedAdamW
+from vit_10b_fsdp_example_amd.parallel import CommContext, FullyShardedDataParallel
+
+
+@pytest.fixture(autouse=True)
+def _fresh_comm():
+    CommContext.reset()
+    yield
+    CommContext.reset()
+
+
+def _tiny_fsdp(seed=0):
+    torch.manual_seed(seed)
+    m = nn.Sequential(nn.Linear(8, 16), nn.GELU(), nn.Linear(16, 4))
+    return FullyShardedDataParallel(m, compute_dtype=torch.float32)
+
+
+def _step_data(seed=1):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(4, 8, generator=g), torch.randn(4, 4, generator=g)
+
+
+def test_comm_grad_attached_and_consumed():
+    fsdp = _tiny_fsdp()
+    x, y = _step_data()
+    (fsdp(x) - y).pow(2).sum().backward()
+    p = fsdp.flat_param
+    assert p.grad is None
+    assert p._comm_grad is not None
+    assert p._grad_prescale == 1.0  # ws=1
+    opt = FusedAdamW(fsdp.parameters(), lr=1e-3, weight_decay=0.1)
+    opt.step()
+    assert p._comm_grad is None  # consumed
+
+
+def test_comm_grad_matches_fp32_ingest_trajectory():
+    results = []
+    for env in ("0", "1"):
+        os.environ["VITFSDP_FP32_GRAD_INGEST"] = env
+        try:
+            CommContext.reset()
+            fsdp = _tiny_fsdp(seed=7)
+            opt = FusedAdamW(fsdp.parameters(), lr=1e-2, weight_decay=0.1)
+            losses = []
+            for step in range(4):
+                x, y = _step_data(seed=100 + step)
+                loss = (fsdp(x) - y).pow(2).sum()
+                loss.backward()
+                fsdp.clip_grad_norm_(1.0)
+                opt.step()
+                opt.zero_grad(set_to_none=True)
+                losses.append(loss.item())
+            results.append(losses)
+        finally:
+            del os.environ["VITFSDP_FP32_GRAD_INGEST"]
+    assert results[0] == pytest.approx(results[1], rel=1e-6)
+
+
+def test_fp32_ingest_env_installs_dot_grad():
+    os.environ["VITFSDP_FP32_GRAD_INGEST"] = "1"
+    try:
+        fsdp = _tiny_fsdp()
+        x, y = _step_data()
+        (fsdp(x) - y).pow(2).sum().backward()
+        p = fsdp.flat_param
+        assert p.grad is not None and p.grad.dtype == torch.float32
+        assert getattr(p, "_comm_grad", None) is None
+    finally:
+        del os.environ["VITFSDP_FP32_GRAD_INGEST"]
+
+
+def test_zero_grad_discards_pending_comm_grad():
+    fsdp = _tiny_fsdp()
+    opt = FusedAdamW(fsdp.parameters(), lr=1e-3)
+    x, y = _step_data()
+    (fsdp(x) - y).pow(2).sum().backward()
+    assert fsdp.flat_param._comm_grad is not None
+    opt.zero_grad(set_to_none=True)
+    assert fsdp.flat_param._comm_grad is None
+    # a following step with no gradient must be a clean no-op
+    before = fsdp.flat_param.detach().clone()
+    opt.step()
+    assert torch.equal(before, fsdp.flat_param.detach())
+
+
+def test_grad_accumulation_across_backwards():
+    """Two backwards before one step must sum their gradients (the ws=1
+    buffer-aliasing path detaches the pending grad before reuse)."""
+    torch.manual_seed(3)
+    ref = nn.Sequential(nn.Linear(8, 16), nn.GELU(), nn.Linear(16, 4))
+    torch.manual_seed(3)
+    fsdp = FullyShardedDataParallel(
+        nn.Sequential(nn.Linear(8, 16), nn.GELU(), nn.Linear(16, 4)),
+        compute_dtype=torch.float32,
+    )
+    for i in range(2):
+        x, y = _step_data(seed=40 + i)
+        (ref(x) - y).pow(2).sum().backward()
+        (fsdp(x) - y).pow(2).sum().backward()
+    ref_flat = torch.cat([p.grad.reshape(-1) for p in ref.parameters()])
+    got = fsdp.flat_param._comm_grad
+    # engine flat order matches construction order here (Linear, Linear)
+    assert got is not None
+    assert torch.allclose(got.sort().values, ref_flat.sort().values, atol=1e-5)
+
+
+def test_fused_adamw_step_divergence_raises():
+    a = nn.Parameter(torch.randn(4))
+    b = nn.Parameter(torch.randn(4))
+    opt = FusedAdamW([a, b], lr=1e-3)
+    a.grad = torch.randn(4)
+    b.grad = torch.randn(4)
+    opt.step()
+    opt.zero_grad(set_to_none=True)
+    a.grad = torch.randn(4)  # b has no grad this step
+    opt.step()
+    opt.zero_grad(set_to_none=True)
+    a.grad = torch.randn(4)
+    b.grad = torch.randn(4)
+    with pytest.raises(RuntimeError, match="diverging"):
+        opt.step()

@@ -225,3 +225,51 @@ def test_shard_on_cpu_training_step():
         losses.append(float(loss.detach()))
     assert all(l == l for l in losses), losses
     assert losses[-1] < losses[0], losses
+
+
+def test_ws1_alias_and_deferred_clip_step():
+    """ws=1 short-circuit: the 'gathered full params' must alias the
+    bf16 mirror storage (no gather copies), the reduced grad shard must
+    be the flat grad buffer itself, and a full deferred-clip training
+    step through that path must track the fp32 eager reference."""
+    from vit_10b_fsdp_example_amd.cli import parse_args
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.models import build_fsdp_vit_model
+    from vit_10b_fsdp_example_amd.ops import CrossEntropyLoss, FusedAdamW
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    CommContext.reset()
+    cfg = parse_args([
+        "--fake_data", "--image_size", "224", "--patch_size", "14",
+        "--embed_dim", "640", "--num_heads", "4", "--num_blocks", "2",
+        "--num_classes", "100", "--batch_size", "8", "--num_workers", "0",
+    ])
+    device = xdist.init_distributed()
+    torch.manual_seed(0)
+    model = build_fsdp_vit_model(cfg, device, compute_dtype=torch.bfloat16)
+    units = model._all_units()
+    for u in units:
+        assert u._ws1_alias, "ws=1 bf16 unit must alias the mirror"
+        assert (
+            u._full_flat.untyped_storage().data_ptr()
+            == u._mirror.untyped_storage().data_ptr()
+        )
+    loss_fn = CrossEntropyLoss()
+    opt = FusedAdamW(model.parameters(), lr=3e-3, weight_decay=0.1)
+    x = torch.randn(8, 3, 224, 224, device=device, dtype=torch.bfloat16)
+    y = torch.randint(0, 100, (8,), device=device)
+    losses = []
+    for _ in range(8):
+        loss = loss_fn(model(x), y)
+        loss.backward()
+        # grads must be the engine's own flat buffers (zero-copy handoff)
+        for u in units:
+            cg = u.flat_param._comm_grad
+            assert cg is not None and cg.dtype == torch.bfloat16
+            assert cg.data_ptr() == u._full_grad.data_ptr()
+        model.clip_grad_norm_(1.0, defer_scale=True)
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        losses.append(float(loss.detach()))
+    assert all(l == l for l in losses), f"NaN loss: {losses}"
+    assert losses[-1] < losses[0] - 0.3, f"loss not decreasing: {losses}"

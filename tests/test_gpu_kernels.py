@@ -414,3 +414,53 @@ def test_lt_gemm_matches_matmul():
     refb = ref + bias.float()
     np.testing.assert_allclose(db.float().cpu().numpy(), refb.cpu().numpy(),
                                atol=0.02 * scale, rtol=0.05)
+
+
+def test_fused_adamw_bf16_comm_grads_prescale():
+    """The engine's comm-dtype handoff: bf16 _comm_grad + _grad_prescale
+    (+ deferred clip scale) must match torch.optim.AdamW fed the
+    explicitly scaled fp32 gradient (fsdp.py _finalize_unit contract)."""
+    torch.manual_seed(1)
+    n = 65539  # scalar-tail exercise
+    from vit_10b_fsdp_example_amd.ops import FusedAdamW
+
+    base = torch.randn(n, device=_dev())
+    g_bf16 = torch.randn(n, device=_dev()).to(torch.bfloat16)
+    prescale = 1.0 / 8.0
+    clip = torch.tensor(0.37, device=_dev())
+
+    a = torch.nn.Parameter(base.clone())
+    a._comm_grad = g_bf16.clone()
+    a._grad_prescale = prescale
+    a._deferred_grad_scale = clip
+    opt1 = FusedAdamW([a], lr=1e-3, weight_decay=0.1)
+    opt1.step()
+    assert a._comm_grad is None  # consumed
+
+    b = torch.nn.Parameter(base.clone())
+    b.grad = g_bf16.to(torch.float32) * (prescale * float(clip))
+    opt2 = torch.optim.AdamW([b], lr=1e-3, weight_decay=0.1)
+    opt2.step()
+    np.testing.assert_allclose(
+        a.detach().cpu().numpy(), b.detach().cpu().numpy(), rtol=1e-5, atol=1e-6
+    )
+
+
+def test_multi_tensor_sqnorm_scale_bf16():
+    from vit_10b_fsdp_example_amd.ops import local_sqnorm, scale_
+
+    torch.manual_seed(2)
+    ts = [
+        torch.randn(1000 + i * 9, device=_dev()).to(torch.bfloat16)
+        for i in range(4)
+    ] + [torch.randn(777, device=_dev())]  # mixed fp32+bf16 list
+    ref = sum(float(t.float().pow(2).sum()) for t in ts)
+    got = float(local_sqnorm(ts))
+    assert abs(got - ref) / ref < 1e-4
+    refs = [(t.float() * 0.25).to(t.dtype).clone() for t in ts]
+    scale_(ts, torch.tensor(0.25, device=_dev()))
+    for t, r in zip(ts, refs):
+        np.testing.assert_allclose(
+            t.float().cpu().numpy(), r.float().cpu().numpy(), rtol=1e-2,
+            atol=1e-3,
+        )

@@ -4,7 +4,10 @@ from .attention import attention, attention_qkv, math_attention
 from .cross_entropy import CrossEntropyLoss, cross_entropy
 from .adamw import FusedAdamW
 from .multi_tensor import local_sqnorm, scale_
-from .linear import NativeLinear, NativeWgradMode, wgrad_backward_context
+from .linear import (
+    NativeLinear, NativeWgradMode, TunedGemmMode, gemm_dispatch_context,
+    wgrad_backward_context,
+)
 
 __all__ = [
     "ext",
@@ -23,5 +26,7 @@ __all__ = [
     "scale_",
     "NativeLinear",
     "NativeWgradMode",
+    "TunedGemmMode",
+    "gemm_dispatch_context",
     "wgrad_backward_context",
 ]

@@ -63,6 +63,7 @@ import torch.nn.functional as F
 from torch.utils._python_dispatch import TorchDispatchMode
 
 from ._extension import ext, use_hip
+from ..tuning import lt_algo_table
 
 
 # "1": custom autograd.Function (kept for A/B history; defeats checkpoint
@@ -199,12 +200,122 @@ class NativeWgradMode(TorchDispatchMode):
         return func(*args, **kwargs)
 
 
-def wgrad_backward_context():
-    """Context manager for ``loss.backward()``: NativeWgradMode when
-    VITFSDP_NATIVE_WGRAD=2, otherwise a no-op."""
-    if _NATIVE_WGRAD_DISPATCH:
-        return NativeWgradMode()
+def _op_layout(t):
+    """'N' for a row-contiguous 2-D tensor, 'T' for a transposed view of
+    one, None otherwise (mirrors csrc/ltgemm.cpp classify())."""
+    if t.dim() != 2:
+        return None
+    r, c = t.shape
+    s0, s1 = t.stride()
+    if s1 == 1 and (s0 == c or r == 1):
+        return "N"
+    if s0 == 1 and (s1 == r or c == 1):
+        return "T"
+    return None
+
+
+def _dual_key(a, b):
+    """Column-major-dual shape key for row-major a @ b, matching the
+    offline search rows (csrc/tools/hipblaslt_search.cpp): the A-slot
+    takes b's memory, the B-slot a's."""
+    la, lb = _op_layout(a), _op_layout(b)
+    if la is None or lb is None:
+        return None
+    return (lb, la, b.shape[1], a.shape[0], a.shape[1])
+
+
+class TunedGemmMode(TorchDispatchMode):
+    """Reroute training GEMMs whose shape has an offline-searched
+    hipBLASLt algorithm (tuned/lt_algos_gfx950.json) through
+    _C.lt_gemm with the explicit index — same library, better kernel
+    pick than the heuristic (ROADMAP item 5).  Dispatcher-level like
+    NativeWgradMode, so the autograd graph keeps its stock addmm/mm
+    nodes and non-reentrant checkpoint early-stop is preserved.
+
+    Also hosts the native-wgrad reroute when VITFSDP_NATIVE_WGRAD=2 so
+    one mode covers the whole step (push around forward AND backward).
+    """
+
+    def __init__(self, table=None, native_wgrad=None, handler=None):
+        super().__init__()
+        self.table = lt_algo_table() if table is None else table
+        self.native_wgrad = (
+            _NATIVE_WGRAD_DISPATCH if native_wgrad is None else native_wgrad
+        )
+        # test seam: handler(a, b, idx, bias) replaces _C.lt_gemm (and
+        # lifts the CUDA/bf16 gate) so the routing logic runs on CPU
+        self._handler = handler
+        self.hits = 0
+        self.wgrad_hits = 0
+
+    def _tuned_index(self, a, b):
+        if not self.table:
+            return None
+        key = _dual_key(a, b)
+        if key is None:
+            return None
+        return self.table.get(key)
+
+    def _gpu_ok(self, t):
+        if self._handler is not None:
+            return True
+        return t.is_cuda and t.dtype == torch.bfloat16 and use_hip(t)
+
+    def _route(self, a, b, idx, bias=None):
+        if self._handler is not None:
+            return self._handler(a, b, idx, bias)
+        return ext().lt_gemm(a, b, idx, bias)
+
+    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
+        kwargs = kwargs or {}
+        if func is torch.ops.aten.mm.default and self._gpu_ok(args[0]):
+            a, b = args
+            if (
+                self.native_wgrad
+                and self._handler is None
+                and _is_wgrad_mm(a, b)
+                and hasattr(ext(), "wgrad_gemm")
+            ):
+                self.wgrad_hits += 1
+                (dw,) = ext().wgrad_gemm(a.t().contiguous(), b.contiguous(),
+                                         False)
+                return dw
+            idx = self._tuned_index(a, b)
+            if idx is not None:
+                self.hits += 1
+                return self._route(a, b, idx)
+        elif (
+            func is torch.ops.aten.addmm.default
+            and len(args) == 3
+            and kwargs.get("beta", 1) == 1
+            and kwargs.get("alpha", 1) == 1
+            and self._gpu_ok(args[1])
+            and args[0].dim() == 1
+        ):
+            bias, a, b = args
+            idx = self._tuned_index(a, b)
+            if idx is not None:
+                self.hits += 1
+                return self._route(a, b, idx, bias)
+        return func(*args, **kwargs)
+
+
+def gemm_dispatch_context():
+    """Context manager for the training step (forward and backward):
+    activates the dispatcher-level GEMM rerouting when either the tuned
+    algorithm table is present or VITFSDP_NATIVE_WGRAD=2; otherwise a
+    no-op."""
+    if _NATIVE_WGRAD_DISPATCH and not lt_algo_table():
+        return NativeWgradMode()  # backward-compatible, wgrad only
+    if lt_algo_table():
+        return TunedGemmMode()
     return contextlib.nullcontext()
+
+
+def wgrad_backward_context():
+    """Backward-compatible alias of gemm_dispatch_context (historical
+    name from when only the dW GEMMs were rerouted)."""
+    return gemm_dispatch_context()
 
 
 class NativeLinear(nn.Linear):

@@ -72,7 +72,7 @@ class FullyShardedDataParallel(nn.Module):
         self,
         module,
         reshard_after_forward=True,
-        flatten_parameters=True,
+        flatten_parameters=False,
         compute_dtype=torch.float32,
         device=None,
         prefetch=True,
@@ -83,17 +83,18 @@ class FullyShardedDataParallel(nn.Module):
         # reference (run_vit_training.py:180,359).  This engine is
         # flat-by-design: a unit's parameters always live in one flat
         # buffer, because one large RCCL message per unit is the xGMI-
-        # efficient shape.  The flag therefore has no effect (documented
-        # deviation; consolidation understands our layout either way) —
-        # warn once so passing it is never a silent behavior change.
+        # efficient shape.  The flag therefore has no effect either way
+        # (documented deviation; consolidation understands our layout) —
+        # warn once when --flatten_parameters is requested so the flag is
+        # never a silent behavior change.
         if flatten_parameters and not FullyShardedDataParallel._flat_warned:
             FullyShardedDataParallel._flat_warned = True
             import warnings
 
             warnings.warn(
-                "flatten_parameters is accepted for reference-CLI "
-                "compatibility but has no effect: this FSDP engine is "
-                "flat-by-design (one flat buffer per unit)",
+                "--flatten_parameters accepted for reference-CLI "
+                "compatibility: this FSDP engine is flat-by-design (one "
+                "flat buffer per unit), so the flag changes nothing",
                 stacklevel=2,
             )
         self.reshard_after_forward = reshard_after_forward

@@ -23,7 +23,7 @@ import torch
 from . import dist as xdist
 from .data import build_datasets
 from .models import build_fsdp_vit_model
-from .ops import CrossEntropyLoss, FusedAdamW, wgrad_backward_context
+from .ops import CrossEntropyLoss, FusedAdamW, gemm_dispatch_context
 from .utils import SmoothedValue, get_warmup_cosine_scheduler, save_ckpt, load_ckpt
 
 MODEL_SEED = 1234
@@ -126,13 +126,12 @@ def train(cfg):
         for step, (data, target) in enumerate(train_loader):
             if max_steps and step >= max_steps:
                 break
-            # 1. forward
-            output = model(data)
-            loss = loss_fn(output, target)
-
-            # 2. backward + clipping (wgrad_backward_context is a no-op
-            # unless VITFSDP_NATIVE_WGRAD=2 routes dW to csrc/wgemm.hip)
-            with wgrad_backward_context():
+            # 1-2. forward + backward under the GEMM dispatch router
+            # (no-op unless a tuned hipBLASLt table is present or
+            # VITFSDP_NATIVE_WGRAD=2 routes dW to csrc/wgemm.hip)
+            with gemm_dispatch_context():
+                output = model(data)
+                loss = loss_fn(output, target)
                 loss.backward()
             if not cfg.run_without_fsdp:
                 # clip on the FULL (not per-shard) gradient norm — the

@@ -46,3 +46,44 @@ def enable_tunableop():
             torch.cuda.tunable.enable(False)
         except Exception:
             pass
+
+
+# ---------------------------------------------------------------------------
+# Offline hipBLASLt algorithm table (csrc/tools/hipblaslt_search.cpp ->
+# scripts/make_lt_table.py -> tuned/lt_algos_gfx950.json).  Keys are the
+# column-major dual of each training GEMM, "opA,opB,m,n,k"; values the
+# library algorithm index that beat the heuristic by the gate margin.
+# The table is measured data: committing it IS the default flip (the
+# dispatch router activates whenever a table is present).
+# ---------------------------------------------------------------------------
+
+_LT_TABLE = None
+
+
+def lt_algo_table():
+    """{(opA, opB, m, n, k): algo_index} from tuned/lt_algos_gfx950.json,
+    or {} when absent / disabled via VITFSDP_TUNED_GEMM=0."""
+    global _LT_TABLE
+    if _LT_TABLE is not None:
+        return _LT_TABLE
+    if os.environ.get("VITFSDP_TUNED_GEMM", "1") == "0":
+        _LT_TABLE = {}
+        return _LT_TABLE
+    path = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "tuned", "lt_algos_gfx950.json",
+    )
+    table = {}
+    if os.path.exists(path):
+        import json
+
+        try:
+            raw = json.load(open(path))
+            for key, val in raw.get("entries", {}).items():
+                opa, opb, m, n, k = key.split(",")
+                table[(opa, opb, int(m), int(n), int(k))] = int(val["index"])
+        except Exception as exc:  # pragma: no cover
+            print(f"[tuning] lt algo table unreadable ({exc!r})", flush=True)
+            table = {}
+    _LT_TABLE = table
+    return _LT_TABLE
