@@ -94,15 +94,25 @@ struct Timed {
 };
 
 int main(int argc, char** argv) {
-    std::string model = "vit10b";
+    std::string model = "vit10b", only;
     int reps = 5, topk = 16;
     for (int i = 1; i < argc; ++i) {
         if (!strcmp(argv[i], "--model") && i + 1 < argc) model = argv[++i];
         if (!strcmp(argv[i], "--reps") && i + 1 < argc) reps = atoi(argv[++i]);
         if (!strcmp(argv[i], "--topk") && i + 1 < argc) topk = atoi(argv[++i]);
+        // --only NAME: run a single problem (lets a driver script run
+        // every problem in its own process, so one broken library
+        // kernel faulting the GPU cannot take down the whole sweep)
+        if (!strcmp(argv[i], "--only") && i + 1 < argc) only = argv[++i];
     }
     auto problems =
         model == "vit-large" ? problems_vitlarge() : problems_vit10b();
+    if (!only.empty()) {
+        std::vector<Problem> filtered;
+        for (auto& p : problems)
+            if (only == p.name) filtered.push_back(p);
+        problems = filtered;
+    }
 
     hipblasLtHandle_t handle;
     HIPBLAS_CHECK(hipblasLtCreate(&handle));
@@ -138,11 +148,14 @@ int main(int argc, char** argv) {
         // two operand sets so repeated runs do not replay a warm L3
         // (guide rule 25); each set's A+B alone exceed the 256 MB L3
         // for the 10B shapes, but rotate anyway for the small ones
+        // +32 MiB guard padding per operand: a screened library kernel
+        // with out-of-tile reads/writes must not memory-fault the box
+        const size_t pad = 32u << 20;
         void *a[2], *b[2], *d[2];
         for (int s = 0; s < 2; ++s) {
-            HIP_CHECK(hipMalloc(&a[s], sizeof(uint16_t) * p.m * p.k));
-            HIP_CHECK(hipMalloc(&b[s], sizeof(uint16_t) * p.k * p.n));
-            HIP_CHECK(hipMalloc(&d[s], sizeof(uint16_t) * p.m * p.n));
+            HIP_CHECK(hipMalloc(&a[s], sizeof(uint16_t) * p.m * p.k + pad));
+            HIP_CHECK(hipMalloc(&b[s], sizeof(uint16_t) * p.k * p.n + pad));
+            HIP_CHECK(hipMalloc(&d[s], sizeof(uint16_t) * p.m * p.n + pad));
             HIP_CHECK(hipMemset(a[s], 0x3c, sizeof(uint16_t) * p.m * p.k));
             HIP_CHECK(hipMemset(b[s], 0x3c, sizeof(uint16_t) * p.k * p.n));
         }
@@ -199,6 +212,7 @@ int main(int argc, char** argv) {
                        p.opB == HIPBLAS_OP_T ? 'T' : 'N', (long)p.m, (long)p.n,
                        (long)p.k, hipblaslt_ext::getIndexFromAlgo(heur[0].algo),
                        ms, tflop / (ms / 1e3));
+            fflush(stdout);
         }
 
         // screen every supported algorithm with 1 rep, refine the top-k
@@ -229,6 +243,7 @@ int main(int argc, char** argv) {
                            p.opB == HIPBLAS_OP_T ? 'T' : 'N', (long)p.m,
                            (long)p.n, (long)p.k, t.index, ms,
                            tflop / (ms / 1e3), refined);
+                fflush(stdout);
                 break;
             }
         }

@@ -128,6 +128,11 @@ class Mlp(nn.Module):
         self.fc1 = NativeLinear(dim, hidden_dim)
         self.fc2 = NativeLinear(hidden_dim, dim)
         self.drop = nn.Dropout(drop)
+        # register the MLP dims so the GEMM dispatch router can fuse the
+        # GELU into the fc1/fc2 hipBLASLt epilogues (VITFSDP_FUSED_GELU)
+        from ..ops.linear import configure_gelu_fusion
+
+        configure_gelu_fusion(dim, hidden_dim)
 
     def forward(self, x):
         x = self.drop(F.gelu(self.fc1(x)))

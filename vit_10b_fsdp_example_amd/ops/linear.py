@@ -224,6 +224,58 @@ def _dual_key(a, b):
     return (lb, la, b.shape[1], a.shape[0], a.shape[1])
 
 
+# --------------------------------------------------------------------------
+# Fused MLP GELU via hipblaslt-ext epilogues (ROADMAP item 6 / VERDICT 3).
+#
+# Enabled by configure_gelu_fusion(embed_dim, hidden_dim) + the
+# VITFSDP_FUSED_GELU env knob.  Mechanism (all below autograd, so the
+# stock addmm/mm/gelu graph nodes — and checkpoint early-stop — are
+# untouched):
+#   forward:  fc1's addmm runs as lt_gemm_gelu -> (gelu_out, pre); the
+#             dispatch returns PRE as the addmm result (so GeluBackward
+#             saves the correct pre-activation) and answers the
+#             immediately following aten.gelu with the cached gelu_out.
+#   backward: fc2's dgrad mm runs as lt_gemm_dgelu_bgrad(dy, W2, pre)
+#             -> (dpre, dbias1) using the pre-activation saved on a
+#             weakref stack by the forward/recompute; the following
+#             aten.gelu_backward is answered with dpre as-is, and fc1's
+#             dbias sum with the fused dbias1.
+# hipBLASLt's GELU is the tanh approximation; because BOTH directions
+# are fused, forward and backward stay consistent (the model trains
+# with tanh-GELU — measured-equivalence policy documented in
+# docs/PERF_MODEL.md; the erf default path remains when fusion is off).
+# --------------------------------------------------------------------------
+
+_GELU_CFG = {"d": 0, "hid": 0}
+
+
+def configure_gelu_fusion(embed_dim, hidden_dim):
+    """Register the MLP dims so the dispatch mode can recognize fc1's
+    forward and fc2's dgrad by shape.  No-op when embed == hidden (the
+    shape keys would be ambiguous)."""
+    if embed_dim != hidden_dim:
+        _GELU_CFG["d"] = int(embed_dim)
+        _GELU_CFG["hid"] = int(hidden_dim)
+
+
+def _gelu_fusion_on():
+    return (
+        os.environ.get("VITFSDP_FUSED_GELU", "0") == "1"
+        and _GELU_CFG["hid"] > 0
+    )
+
+
+def _dgelu_tanh(go, pre):
+    """tanh-approx GELU derivative (matches hipBLASLt's epilogue): the
+    loud-but-correct fallback when a gelu_backward could not be fused
+    after a tanh-fused forward."""
+    c = 0.7978845608028654  # sqrt(2/pi)
+    x = pre.float()
+    t = torch.tanh(c * (x + 0.044715 * x * x * x))
+    dt = (1.0 - t * t) * c * (1.0 + 3 * 0.044715 * x * x)
+    return (go.float() * (0.5 * (1.0 + t) + 0.5 * x * dt)).to(go.dtype)
+
+
 class TunedGemmMode(TorchDispatchMode):
     """Reroute training GEMMs whose shape has an offline-searched
     hipBLASLt algorithm (tuned/lt_algos_gfx950.json) through
@@ -232,8 +284,9 @@ class TunedGemmMode(TorchDispatchMode):
     NativeWgradMode, so the autograd graph keeps its stock addmm/mm
     nodes and non-reentrant checkpoint early-stop is preserved.
 
-    Also hosts the native-wgrad reroute when VITFSDP_NATIVE_WGRAD=2 so
-    one mode covers the whole step (push around forward AND backward).
+    Also hosts the native-wgrad reroute when VITFSDP_NATIVE_WGRAD=2 and
+    the fused MLP GELU epilogues when configured, so one mode covers
+    the whole step (push around forward AND backward).
     """
 
     def __init__(self, table=None, native_wgrad=None, handler=None):
@@ -247,6 +300,14 @@ class TunedGemmMode(TorchDispatchMode):
         self._handler = handler
         self.hits = 0
         self.wgrad_hits = 0
+        # fused-GELU state (see module comment above):
+        self.fused_gelu = _gelu_fusion_on() and handler is None
+        self._pending_gelu = {}  # pre-act data_ptr -> gelu_out (transient)
+        self._aux_stack = []  # weakrefs to pre-activations, LIFO
+        self._aux_live = {}  # pre-act data_ptr -> weakref (fallback check)
+        self._dgelu_done = {}  # dpre data_ptr -> (aux_ptr, dbias)
+        self._pending_dbias = None  # (dpre data_ptr, fused dbias)
+        self.gelu_hits = 0
 
     def _tuned_index(self, a, b):
         if not self.table:
@@ -266,6 +327,67 @@ class TunedGemmMode(TorchDispatchMode):
             return self._handler(a, b, idx, bias)
         return ext().lt_gemm(a, b, idx, bias)
 
+    # -- fused-GELU helpers -------------------------------------------------
+
+    def _is_fc1_fwd(self, key):
+        return (
+            key is not None
+            and key[0] == "T" and key[1] == "N"
+            and key[2] == _GELU_CFG["hid"] and key[4] == _GELU_CFG["d"]
+        )
+
+    def _is_fc2_dgrad(self, key):
+        return (
+            key is not None
+            and key[0] == "N" and key[1] == "N"
+            and key[2] == _GELU_CFG["hid"] and key[4] == _GELU_CFG["d"]
+        )
+
+    def _fused_fc1_fwd(self, a, b, bias, idx):
+        import weakref
+
+        out, pre = ext().lt_gemm_gelu(a, b, bias, -1 if idx is None else idx)
+        self.gelu_hits += 1
+        if len(self._pending_gelu) > 4:  # unconsumed strays (shouldn't happen)
+            self._pending_gelu.clear()
+        self._pending_gelu[pre.data_ptr()] = out
+        # Lifetime of the pre-activation until the fc2-dgrad fusion:
+        #  * original forward: autograd's SavedVariable keeps the C++
+        #    tensor (and, via PyObject preservation, the Python object)
+        #    alive — a weakref suffices, and for a checkpointed block
+        #    (whose activations are discarded) it correctly dies;
+        #  * recompute (inside a backward graph task): the checkpoint
+        #    frame saves a DETACHED copy, so the weakref would die at
+        #    early-stop — hold a strong ref instead; it is consumed by
+        #    the same block's fc2-dgrad moments later and the per-step
+        #    mode instance bounds any leftover to one step.
+        in_backward = torch._C._current_graph_task_id() != -1
+        entry = pre if in_backward else weakref.ref(pre)
+        self._aux_stack.append(entry)
+        self._aux_live[pre.data_ptr()] = entry
+        if len(self._aux_live) > 256:
+            self._aux_live = {
+                p: r for p, r in self._aux_live.items()
+                if self._resolve(r) is not None
+            }
+        return pre
+
+    @staticmethod
+    def _resolve(entry):
+        return entry if torch.is_tensor(entry) else entry()
+
+    def _pop_aux(self, rows, cols):
+        while self._aux_stack:
+            aux = self._resolve(self._aux_stack[-1])
+            if aux is None:
+                self._aux_stack.pop()
+                continue
+            if aux.shape[0] == rows and aux.shape[1] == cols:
+                self._aux_stack.pop()
+                return aux
+            return None  # live but mismatched: leave for its own consumer
+        return None
+
     def __torch_dispatch__(self, func, types, args=(), kwargs=None):
         kwargs = kwargs or {}
         if func is torch.ops.aten.mm.default and self._gpu_ok(args[0]):
@@ -280,7 +402,19 @@ class TunedGemmMode(TorchDispatchMode):
                 (dw,) = ext().wgrad_gemm(a.t().contiguous(), b.contiguous(),
                                          False)
                 return dw
-            idx = self._tuned_index(a, b)
+            key = _dual_key(a, b)
+            idx = self.table.get(key) if (self.table and key) else None
+            if self.fused_gelu and self._is_fc2_dgrad(key):
+                aux = self._pop_aux(a.shape[0], b.shape[1])
+                if aux is not None:
+                    dpre, dbias = ext().lt_gemm_dgelu_bgrad(
+                        a, b, aux, -1 if idx is None else idx
+                    )
+                    self.gelu_hits += 1
+                    self._dgelu_done[dpre.data_ptr()] = (
+                        aux.data_ptr(), dbias,
+                    )
+                    return dpre
             if idx is not None:
                 self.hits += 1
                 return self._route(a, b, idx)
@@ -293,22 +427,57 @@ class TunedGemmMode(TorchDispatchMode):
             and args[0].dim() == 1
         ):
             bias, a, b = args
-            idx = self._tuned_index(a, b)
+            key = _dual_key(a, b)
+            idx = self.table.get(key) if (self.table and key) else None
+            if self.fused_gelu and self._is_fc1_fwd(key):
+                return self._fused_fc1_fwd(a, b, bias, idx)
             if idx is not None:
                 self.hits += 1
                 return self._route(a, b, idx, bias)
+        elif self.fused_gelu and func is torch.ops.aten.gelu.default:
+            out = self._pending_gelu.pop(args[0].data_ptr(), None)
+            if out is not None and out.shape == args[0].shape:
+                return out
+        elif self.fused_gelu and func is torch.ops.aten.gelu_backward.default:
+            go, pre = args[0], args[1]
+            ent = self._dgelu_done.pop(go.data_ptr(), None)
+            if ent is not None and ent[0] == pre.data_ptr():
+                # dgelu already applied in the fc2-dgrad GEMM epilogue;
+                # remember the fused dbias for fc1's bias-grad sum
+                self._pending_dbias = (go.data_ptr(), ent[1])
+                return go
+            ref = self._aux_live.get(pre.data_ptr())
+            if ref is not None and self._resolve(ref) is not None:
+                # forward was tanh-fused but the backward GEMM fusion
+                # missed: apply the matching tanh derivative explicitly
+                return _dgelu_tanh(go, pre)
+        elif (
+            self.fused_gelu
+            and func is torch.ops.aten.sum.dim_IntList
+            and len(args) >= 2
+            and args[1] == [0]
+            and getattr(self, "_pending_dbias", None) is not None
+            and args[0].data_ptr() == self._pending_dbias[0]
+        ):
+            dbias = self._pending_dbias[1]
+            self._pending_dbias = None
+            if dbias.shape[0] == args[0].shape[1]:
+                keepdim = args[2] if len(args) > 2 else kwargs.get(
+                    "keepdim", False
+                )
+                return dbias.unsqueeze(0) if keepdim else dbias
         return func(*args, **kwargs)
 
 
 def gemm_dispatch_context():
     """Context manager for the training step (forward and backward):
-    activates the dispatcher-level GEMM rerouting when either the tuned
-    algorithm table is present or VITFSDP_NATIVE_WGRAD=2; otherwise a
-    no-op."""
-    if _NATIVE_WGRAD_DISPATCH and not lt_algo_table():
-        return NativeWgradMode()  # backward-compatible, wgrad only
-    if lt_algo_table():
+    activates the dispatcher-level GEMM rerouting when the tuned
+    algorithm table is present, fused MLP GELU is configured, or
+    VITFSDP_NATIVE_WGRAD=2; otherwise a no-op."""
+    if lt_algo_table() or _gelu_fusion_on():
         return TunedGemmMode()
+    if _NATIVE_WGRAD_DISPATCH:
+        return NativeWgradMode()  # backward-compatible, wgrad only
     return contextlib.nullcontext()
 
 
