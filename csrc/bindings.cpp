@@ -56,6 +56,8 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
+torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
+                       c10::optional<torch::Tensor> bias);
 torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
                       c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> lt_gemm_gelu(torch::Tensor a, torch::Tensor b,
@@ -100,6 +102,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("wgrad_gemm", &wgrad_gemm,
         "C = A^T B weight-gradient GEMM (bf16, tr16 transpose reads)");
+  m.def("fwd_gemm", &fwd_gemm,
+        "C = X W^T forward Linear GEMM (+fused bias), hand-written CDNA4 "
+        "MFMA (csrc/fgemm.hip)",
+        py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
   m.def("lt_gemm", &lt_gemm,
         "row-major bf16 GEMM (+optional fused bias epilogue) via "
         "hipblaslt-ext with an explicit algorithm index (-1 = library "

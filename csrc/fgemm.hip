@@ -1,0 +1,233 @@
+// Forward Linear GEMM for CDNA4: C[M,N] = A B^T with A = X [M, K] and
+// B = W [N, K] both row-major bf16 (+ optional fused bias over N) — the
+// Linear-forward shape, M = batch*tokens = 32768 for ViT-10B.
+//
+// This is the EASY orientation for MFMA on gfx950: the contraction dim
+// K is the contiguous axis of BOTH operands, and the 16x16x32 bf16 MFMA
+// wants each lane to hold 8 CONTIGUOUS k elements of its row/column —
+// so both global staging and LDS fragment reads are plain vectorized
+// b128 accesses (no ds_read_b64_tr_b16 hardware transposes needed,
+// unlike the wgrad kernel csrc/wgemm.hip whose skeleton this reuses:
+// same 256x256x64 tile, 512 threads as 2m x 4n waves of 128x64, double
+// buffered LDS, next-step global loads issued under the MFMA phases).
+//
+// MFMA fragment layout (pinned on-device by mfma_probe,
+// tests/test_gpu_kernels.py): lane l of the wave contributes row (l&15)
+// of A and column (l&15) of B, k-slice (l>>4)*8 .. +8; the f32x4 result
+// holds rows (l>>4)*4 .. +4 of column (l&15).
+
+#ifndef VITFSDP_KERNELS_ONLY
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+#endif
+
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int kBM = 256;
+constexpr int kBN = 256;
+constexpr int kBK = 64;
+constexpr int kThreads = 512;  // 8 waves: 2 (m) x 4 (n)
+// LDS tiles are [row][k] with the k row padded +8 elements (16 B): the
+// row stride is then 144 B, so the 16 lanes of a fragment-read group
+// (consecutive rows, same k column) touch distinct bank lines
+constexpr int kSK = kBK + 8;
+
+struct FgemmShared {
+  short a_tile[2][kBM][kSK];  // X tile, [m][k]
+  short b_tile[2][kBN][kSK];  // W tile, [n][k]
+};
+
+template <bool WITH_BIAS>
+__global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
+    const short* __restrict__ a,     // [M, K]
+    const short* __restrict__ b,     // [N, K]
+    const short* __restrict__ bias,  // optional [N], may be null
+    short* __restrict__ c,           // [M, N]
+    int M, int N, int K) {
+  HIP_DYNAMIC_SHARED(char, smem_raw)
+  FgemmShared& sm = *reinterpret_cast<FgemmShared*>(smem_raw);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int col = lane & 15;  // fragment row (A) / column (B)
+  const int seg = lane >> 4;  // k-slice 8*seg within the 32-k chunk
+  const int wm = wave >> 2;   // 0..1: wave's 128-row m strip
+  const int wn = wave & 3;    // 0..3: wave's 64-col n strip
+
+  // XCD-aware tile swizzle (guide T1, same as wgemm.hip): give each XCD
+  // a contiguous run of tiles so its co-resident blocks share operand
+  // panels in its private L2
+  int bx = blockIdx.x, by = blockIdx.y;
+  {
+    const int nwg = gridDim.x * gridDim.y;
+    if ((nwg & 7) == 0) {
+      const int bid = (int)(blockIdx.x + blockIdx.y * gridDim.x);
+      const int cpx = nwg >> 3;
+      const int swz = (bid & 7) * cpx + (bid >> 3);
+      bx = swz % gridDim.x;
+      by = swz / gridDim.x;
+    }
+  }
+  const long m0 = (long)bx * kBM;
+  const long n0 = (long)by * kBN;
+
+  // staging: thread t loads rows (t>>3) + i*64, k-cols (t&7)*8 (+16 B
+  // vectors).  256 rows x 64 k per tile, 512 threads -> 4 vectors each.
+  constexpr int kVecs = kBM * kBK / 8 / kThreads;  // 4
+  const int s_r0 = tid >> 3;
+  const int s_c8 = (tid & 7) * 8;
+  bf16x8 a_st[kVecs], b_st[kVecs];
+  auto issue_loads = [&](long k_base) {
+#pragma unroll
+    for (int i = 0; i < kVecs; ++i) {
+      const long row = s_r0 + i * (kThreads / 8);
+      a_st[i] =
+          *reinterpret_cast<const bf16x8*>(&a[(m0 + row) * K + k_base + s_c8]);
+      b_st[i] =
+          *reinterpret_cast<const bf16x8*>(&b[(n0 + row) * K + k_base + s_c8]);
+    }
+  };
+  auto write_tiles = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < kVecs; ++i) {
+      const int row = s_r0 + i * (kThreads / 8);
+      *reinterpret_cast<bf16x8*>(&sm.a_tile[buf][row][s_c8]) = a_st[i];
+      *reinterpret_cast<bf16x8*>(&sm.b_tile[buf][row][s_c8]) = b_st[i];
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_ksteps = K / kBK;
+  issue_loads(0);
+  write_tiles(0);
+  __syncthreads();
+
+  for (int ks = 0; ks < n_ksteps; ++ks) {
+    const int buf = ks & 1;
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      // next K-step's global loads issue at the start of the second
+      // chunk: latency hides under its MFMAs (wgemm.hip T14 scheme)
+      if (kc == 1 && ks + 1 < n_ksteps) issue_loads((long)(ks + 1) * kBK);
+      const int kcol = kc * 32 + seg * 8;
+      // B fragments for this chunk (4 plain b128 LDS reads, reused
+      // across all 8 m fragments)
+      bf16x8 bf[4];
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            &sm.b_tile[buf][wn * 64 + ni * 16 + col][kcol]);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
+        const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+            &sm.a_tile[buf][wm * 128 + mi * 16 + col][kcol]);
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni) {
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af, bf[ni], acc[mi][ni], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (ks + 1 < n_ksteps) write_tiles(buf ^ 1);
+    __syncthreads();
+  }
+
+  // epilogue: this lane owns rows m = ... + seg*4 + r of column n
+  float bias_v[4];
+  if (WITH_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      bias_v[ni] = bf16_to_f32(
+          (unsigned short)bias[n0 + wn * 64 + ni * 16 + col]);
+    }
+  }
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long m = m0 + wm * 128 + mi * 16 + seg * 4 + r;
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const long n = n0 + wn * 64 + ni * 16 + col;
+        float v = acc[mi][ni][r];
+        if (WITH_BIAS) v += bias_v[ni];
+        c[m * N + n] = (short)f32_to_bf16(v);
+      }
+    }
+  }
+}
+
+#ifdef VITFSDP_KERNELS_ONLY
+template __global__ void fgemm_abt_kernel<false>(const short*, const short*,
+                                                 const short*, short*, int,
+                                                 int, int);
+template __global__ void fgemm_abt_kernel<true>(const short*, const short*,
+                                                const short*, short*, int,
+                                                int, int);
+#endif
+
+}  // namespace
+
+#ifndef VITFSDP_KERNELS_ONLY
+torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
+                       c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous(),
+              "fwd_gemm: contiguous CUDA tensors required");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16, "fwd_gemm: bf16 only");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "fwd_gemm: x[M,K], w[N,K] expected");
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % kBK == 0,
+              "fwd_gemm: needs M%256==0, N%256==0, K%64==0 (got ", M, ",", N,
+              ",", K, ")");
+  auto c = torch::empty({M, N}, x.options());
+  const short* bias_ptr = nullptr;
+  torch::Tensor bias_c;
+  if (bias.has_value()) {
+    bias_c = bias->contiguous();
+    TORCH_CHECK(bias_c.scalar_type() == torch::kBFloat16 &&
+                bias_c.numel() == N, "fwd_gemm: bias must be bf16 [N]");
+    bias_ptr = (const short*)bias_c.data_ptr();
+  }
+  static bool attr_set = [] {
+    hipFuncSetAttribute(reinterpret_cast<const void*>(&fgemm_abt_kernel<false>),
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        sizeof(FgemmShared));
+    hipFuncSetAttribute(reinterpret_cast<const void*>(&fgemm_abt_kernel<true>),
+                        hipFuncAttributeMaxDynamicSharedMemorySize,
+                        sizeof(FgemmShared));
+    return true;
+  }();
+  (void)attr_set;
+  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (bias_ptr) {
+    hipLaunchKernelGGL(fgemm_abt_kernel<true>, grid, dim3(kThreads),
+                       sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), bias_ptr,
+                       (short*)c.data_ptr(), (int)M, (int)N, (int)K);
+  } else {
+    hipLaunchKernelGGL(fgemm_abt_kernel<false>, grid, dim3(kThreads),
+                       sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
+                       (const short*)w.data_ptr(), nullptr,
+                       (short*)c.data_ptr(), (int)M, (int)N, (int)K);
+  }
+  HIP_CHECK_LAST();
+  return c;
+}
+#endif  // VITFSDP_KERNELS_ONLY

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         "csrc/cross_entropy.hip",
         "csrc/fmha.hip",
         "csrc/wgemm.hip",
+        "csrc/fgemm.hip",
     ],
     libraries=["hipblaslt"],
     extra_compile_args={

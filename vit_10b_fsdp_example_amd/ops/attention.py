@@ -12,9 +12,30 @@ math composition (matmul + softmax), which is also the numerics
 reference for the kernel tests.
 """
 
+import warnings
+
 import torch
 
 from ._extension import ext, use_hip
+
+_DROPOUT_WARNED = False
+
+
+def _warn_dropout_fallback():
+    """--att_dropout > 0 routes to the O(T^2) math composition (the
+    flash kernel has no in-kernel RNG yet).  The 10B training recipe
+    uses att_dropout 0.0 (reference run_vit_training.py:346 default),
+    so this is off the measured path — but it must never be a silent
+    30x attention slowdown + O(T^2) memory change."""
+    global _DROPOUT_WARNED
+    if not _DROPOUT_WARNED:
+        _DROPOUT_WARNED = True
+        warnings.warn(
+            "attention dropout > 0: falling back from the flash kernel "
+            "to the explicit-math attention path (O(T^2) memory, "
+            "slower). The reference recipe uses --att_dropout 0.",
+            stacklevel=3,
+        )
 
 
 def math_attention(q, k, v, scale=None, dropout_p=0.0, training=False):
@@ -65,6 +86,8 @@ def attention(q, k, v, scale=None, dropout_p=0.0, training=False):
         return _FlashAttentionFn.apply(
             q.contiguous(), k.contiguous(), v.contiguous(), scale
         )
+    if dropout_p > 0.0 and training and use_hip(q):
+        _warn_dropout_fallback()
     return math_attention(q, k, v, scale, dropout_p, training)
 
 
@@ -110,6 +133,8 @@ def attention_qkv(qkv, num_heads, scale=None, dropout_p=0.0, training=False):
     )
     if use_kernel:
         return _FlashAttentionQkvFn.apply(qkv.contiguous(), num_heads, scale)
+    if dropout_p > 0.0 and training and use_hip(qkv):
+        _warn_dropout_fallback()
     q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)
     o = math_attention(q, k, v, scale, dropout_p, training)
     return o.transpose(1, 2).reshape(B, T, H * D)
