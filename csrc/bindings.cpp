@@ -58,6 +58,8 @@ std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
 torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
                        c10::optional<torch::Tensor> bias);
+std::vector<torch::Tensor> fwd_gemm_gelu(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor bias);
 torch::Tensor lt_gemm(torch::Tensor a, torch::Tensor b, long algo_index,
                       c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> lt_gemm_gelu(torch::Tensor a, torch::Tensor b,
@@ -106,6 +108,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "C = X W^T forward Linear GEMM (+fused bias), hand-written CDNA4 "
         "MFMA (csrc/fgemm.hip)",
         py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+  m.def("fwd_gemm_gelu", &fwd_gemm_gelu,
+        "(gelu(X W^T + bias), pre-activation) with the EXACT erf GELU "
+        "fused into the hand-written forward GEMM epilogue",
+        py::arg("x"), py::arg("w"), py::arg("bias"));
   m.def("lt_gemm", &lt_gemm,
         "row-major bf16 GEMM (+optional fused bias epilogue) via "
         "hipblaslt-ext with an explicit algorithm index (-1 = library "

@@ -42,13 +42,20 @@ struct FgemmShared {
   short b_tile[2][kBN][kSK];  // W tile, [n][k]
 };
 
-template <bool WITH_BIAS>
+// epilogue modes: 0 = plain, 1 = +bias, 2 = +bias + exact-erf GELU
+// (writes the pre-activation to aux for the backward — unlike the
+// hipblaslt-ext GELU_AUX_BIAS epilogue this keeps the reference's erf
+// GELU numerics; ~10 extra VALU ops/element disappear under the MFMA
+// phases of a compute-bound tile)
+template <int EPI>
 __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
     const short* __restrict__ a,     // [M, K]
     const short* __restrict__ b,     // [N, K]
     const short* __restrict__ bias,  // optional [N], may be null
     short* __restrict__ c,           // [M, N]
+    short* __restrict__ aux,         // EPI==2: pre-GELU [M, N]
     int M, int N, int K) {
+  constexpr bool WITH_BIAS = EPI >= 1;
   HIP_DYNAMIC_SHARED(char, smem_raw)
   FgemmShared& sm = *reinterpret_cast<FgemmShared*>(smem_raw);
 
@@ -165,6 +172,11 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
         const long n = n0 + wn * 64 + ni * 16 + col;
         float v = acc[mi][ni][r];
         if (WITH_BIAS) v += bias_v[ni];
+        if (EPI == 2) {
+          aux[m * N + n] = (short)f32_to_bf16(v);
+          // exact erf GELU (reference numerics, torch F.gelu default)
+          v = 0.5f * v * (1.f + erff(v * 0.70710678118654752f));
+        }
         c[m * N + n] = (short)f32_to_bf16(v);
       }
     }
@@ -172,12 +184,15 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
 }
 
 #ifdef VITFSDP_KERNELS_ONLY
-template __global__ void fgemm_abt_kernel<false>(const short*, const short*,
-                                                 const short*, short*, int,
-                                                 int, int);
-template __global__ void fgemm_abt_kernel<true>(const short*, const short*,
-                                                const short*, short*, int,
-                                                int, int);
+template __global__ void fgemm_abt_kernel<0>(const short*, const short*,
+                                             const short*, short*, short*,
+                                             int, int, int);
+template __global__ void fgemm_abt_kernel<1>(const short*, const short*,
+                                             const short*, short*, short*,
+                                             int, int, int);
+template __global__ void fgemm_abt_kernel<2>(const short*, const short*,
+                                             const short*, short*, short*,
+                                             int, int, int);
 #endif
 
 }  // namespace
@@ -205,29 +220,58 @@ torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
     bias_ptr = (const short*)bias_c.data_ptr();
   }
   static bool attr_set = [] {
-    hipFuncSetAttribute(reinterpret_cast<const void*>(&fgemm_abt_kernel<false>),
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        sizeof(FgemmShared));
-    hipFuncSetAttribute(reinterpret_cast<const void*>(&fgemm_abt_kernel<true>),
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        sizeof(FgemmShared));
+    for (auto fn : {reinterpret_cast<const void*>(&fgemm_abt_kernel<0>),
+                    reinterpret_cast<const void*>(&fgemm_abt_kernel<1>),
+                    reinterpret_cast<const void*>(&fgemm_abt_kernel<2>)}) {
+      (void)hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                sizeof(FgemmShared));
+    }
     return true;
   }();
   (void)attr_set;
   dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
   auto stream = at::cuda::getCurrentCUDAStream();
   if (bias_ptr) {
-    hipLaunchKernelGGL(fgemm_abt_kernel<true>, grid, dim3(kThreads),
+    hipLaunchKernelGGL(fgemm_abt_kernel<1>, grid, dim3(kThreads),
                        sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
                        (const short*)w.data_ptr(), bias_ptr,
-                       (short*)c.data_ptr(), (int)M, (int)N, (int)K);
+                       (short*)c.data_ptr(), nullptr, (int)M, (int)N, (int)K);
   } else {
-    hipLaunchKernelGGL(fgemm_abt_kernel<false>, grid, dim3(kThreads),
+    hipLaunchKernelGGL(fgemm_abt_kernel<0>, grid, dim3(kThreads),
                        sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
                        (const short*)w.data_ptr(), nullptr,
-                       (short*)c.data_ptr(), (int)M, (int)N, (int)K);
+                       (short*)c.data_ptr(), nullptr, (int)M, (int)N, (int)K);
   }
   HIP_CHECK_LAST();
   return c;
+}
+
+// (gelu(x @ w^T + bias), pre-activation): the erf-exact fused MLP fc1
+// forward — our answer to hipblaslt-ext GELU_AUX_BIAS without the tanh
+// approximation (kernel K6 fusion).
+std::vector<torch::Tensor> fwd_gemm_gelu(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous() &&
+              bias.is_contiguous(), "fwd_gemm_gelu: contiguous CUDA only");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16 &&
+              bias.scalar_type() == torch::kBFloat16,
+              "fwd_gemm_gelu: bf16 only");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1) &&
+              bias.numel() == w.size(0), "fwd_gemm_gelu: x[M,K], w[N,K]");
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % kBK == 0,
+              "fwd_gemm_gelu: needs M%256==0, N%256==0, K%64==0");
+  auto c = torch::empty({M, N}, x.options());
+  auto aux = torch::empty({M, N}, x.options());
+  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(fgemm_abt_kernel<2>, grid, dim3(kThreads),
+                     sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), (const short*)bias.data_ptr(),
+                     (short*)c.data_ptr(), (short*)aux.data_ptr(), (int)M,
+                     (int)N, (int)K);
+  HIP_CHECK_LAST();
+  return {c, aux};
 }
 #endif  // VITFSDP_KERNELS_ONLY

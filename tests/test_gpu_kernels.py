@@ -581,3 +581,19 @@ def test_fwd_gemm_vs_reference():
     ref_b = ref + bias.float()
     rel_b = (got_b.float() - ref_b).abs().max() / ref_b.abs().max()
     assert float(rel_b) < 2e-2, float(rel_b)
+
+
+def test_fwd_gemm_gelu_erf_vs_reference():
+    """fgemm's fused GELU epilogue must match torch's ERF-exact
+    F.gelu(linear(x)) — the reference numerics, unlike hipBLASLt's
+    tanh-approx epilogue."""
+    torch.manual_seed(10)
+    M, N, K = 512, 256, 320
+    x = torch.randn(M, K, device=_dev()).to(torch.bfloat16)
+    w = (torch.randn(N, K, device=_dev()) * 0.05).to(torch.bfloat16)
+    bias = torch.randn(N, device=_dev()).to(torch.bfloat16)
+    out, pre = EXT.fwd_gemm_gelu(x, w, bias)
+    pre_ref = x.float() @ w.float().t() + bias.float()
+    out_ref = torch.nn.functional.gelu(pre_ref)  # erf-exact
+    assert float((pre.float() - pre_ref).abs().max()) < 0.05
+    assert float((out.float() - out_ref).abs().max()) < 0.05
