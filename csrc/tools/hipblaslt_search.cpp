@@ -95,6 +95,7 @@ struct Timed {
 
 int main(int argc, char** argv) {
     std::string model = "vit10b", only;
+    bool with_bias = false;
     int reps = 5, topk = 16;
     for (int i = 1; i < argc; ++i) {
         if (!strcmp(argv[i], "--model") && i + 1 < argc) model = argv[++i];
@@ -104,6 +105,10 @@ int main(int argc, char** argv) {
         // every problem in its own process, so one broken library
         // kernel faulting the GPU cannot take down the whole sweep)
         if (!strcmp(argv[i], "--only") && i + 1 < argc) only = argv[++i];
+        // --bias: search the *_fwd problems WITH the bias epilogue the
+        // training forward actually uses (an index that wins bare can
+        // fail isAlgoSupported once the epilogue is attached)
+        if (!strcmp(argv[i], "--bias")) with_bias = true;
     }
     auto problems =
         model == "vit-large" ? problems_vitlarge() : problems_vit10b();
@@ -160,10 +165,21 @@ int main(int argc, char** argv) {
             HIP_CHECK(hipMemset(b[s], 0x3c, sizeof(uint16_t) * p.k * p.n));
         }
         float alpha = 1.0f, beta = 0.0f;
+        const bool use_bias =
+            with_bias && strstr(p.name, "_fwd") != nullptr;
+        void* bias_buf = nullptr;
+        if (use_bias) {
+            HIP_CHECK(hipMalloc(&bias_buf, sizeof(uint16_t) * p.m + pad));
+            HIP_CHECK(hipMemset(bias_buf, 0x3c, sizeof(uint16_t) * p.m));
+        }
 
         hipblaslt_ext::Gemm gemm(handle, p.opA, p.opB, HIP_R_16BF, HIP_R_16BF,
                                  HIP_R_16BF, HIP_R_16BF, HIPBLAS_COMPUTE_32F);
         hipblaslt_ext::GemmEpilogue epilogue;  // default: no epilogue
+        if (use_bias) {
+            epilogue.setMode(HIPBLASLT_EPILOGUE_BIAS);
+            epilogue.setBiasDataType(HIP_R_16BF);
+        }
         auto set_problem = [&](int s) {
             hipblaslt_ext::GemmInputs inputs;
             inputs.setA(a[s]);
@@ -172,6 +188,7 @@ int main(int argc, char** argv) {
             inputs.setD(d[s]);
             inputs.setAlpha(&alpha);
             inputs.setBeta(&beta);
+            if (use_bias) inputs.setBias(bias_buf);
             HIPBLAS_CHECK(gemm.setProblem(p.m, p.n, p.k, 1, epilogue, inputs));
         };
         set_problem(0);

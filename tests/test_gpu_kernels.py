@@ -482,12 +482,19 @@ def test_lt_gemm_strided_views_match_matmul():
 
 
 def test_lt_gemm_gelu_matches_tanh_reference():
+    """hipblaslt-ext GELU_AUX epilogue — SKIPPED when the library ships
+    no algorithms for it on this stack (probed: ROCm 7.2 / gfx950 has
+    none; the live fusion path is fwd_gemm_gelu instead)."""
     torch.manual_seed(6)
     tok, din, dout = 512, 256, 384
     x = torch.randn(tok, din, device=_dev()).to(torch.bfloat16)
     w = torch.randn(dout, din, device=_dev()).to(torch.bfloat16) * 0.05
     bias = torch.randn(dout, device=_dev()).to(torch.bfloat16)
-    out, pre = EXT.lt_gemm_gelu(x, w.t(), bias, -1)
+    try:
+        out, pre = EXT.lt_gemm_gelu(x, w.t(), bias, -1)
+    except RuntimeError as exc:
+        assert "no heuristic algorithm" in str(exc)
+        pytest.skip("hipBLASLt ships no GELU_AUX algorithms on this stack")
     pre_ref = torch.matmul(x.float(), w.t().float()) + bias.float()
     out_ref = torch.nn.functional.gelu(pre_ref, approximate="tanh")
     assert float((pre.float() - pre_ref).abs().max()) < 0.05
@@ -500,7 +507,11 @@ def test_lt_gemm_dgelu_bgrad_matches_reference():
     dy = torch.randn(tok, dout, device=_dev()).to(torch.bfloat16)
     w2 = torch.randn(dout, hid, device=_dev()).to(torch.bfloat16) * 0.05
     pre = torch.randn(tok, hid, device=_dev()).to(torch.bfloat16)
-    dpre, dbias = EXT.lt_gemm_dgelu_bgrad(dy, w2, pre, -1)
+    try:
+        dpre, dbias = EXT.lt_gemm_dgelu_bgrad(dy, w2, pre, -1)
+    except RuntimeError as exc:
+        assert "no heuristic algorithm" in str(exc)
+        pytest.skip("hipBLASLt ships no DGELU algorithms on this stack")
 
     dgelu_out_ref = torch.matmul(dy.float(), w2.float())
     p = pre.float().detach().requires_grad_(True)
@@ -516,15 +527,17 @@ def test_lt_gemm_dgelu_bgrad_matches_reference():
 
 
 def test_fused_gelu_dispatch_gpu(monkeypatch):
-    """End-to-end dispatch fusion with the real kernels on a small MLP:
-    fused-on must match the eager tanh-GELU composition, under
-    checkpointing, gradients included."""
+    """End-to-end dispatch fusion with the real fwd_gemm_gelu kernel on
+    a small MLP: fused-on must match the STOCK eager erf composition
+    (same numerics up to GEMM rounding), under checkpointing, gradients
+    included."""
     import torch.nn.functional as F
     from torch.utils.checkpoint import checkpoint
     from vit_10b_fsdp_example_amd.ops import linear as linmod
 
     tok, d, hid = 256, 128, 512
     monkeypatch.setenv("VITFSDP_FUSED_GELU", "1")
+    monkeypatch.setattr(linmod, "_FGEMM_MIN_TILES", 0)
     monkeypatch.setitem(linmod._GELU_CFG, "d", d)
     monkeypatch.setitem(linmod._GELU_CFG, "hid", hid)
 
@@ -539,30 +552,25 @@ def test_fused_gelu_dispatch_gpu(monkeypatch):
     def block(t):
         return F.linear(F.gelu(F.linear(t, w1, b1)), w2, b2)
 
-    with linmod.TunedGemmMode() as m:
+    with linmod.TunedGemmMode(table={}, native_wgrad=False) as m:
         y = checkpoint(block, x, use_reentrant=False)
         y.float().pow(2).sum().backward()
-    assert m.gelu_hits == 3, m.gelu_hits  # fwd + recompute + dgelu
+    assert m.gelu_hits == 2, m.gelu_hits  # forward + recompute
 
     grads = [t.grad.clone() for t in (x, w1, b1, w2, b2)]
     for t in (x, w1, b1, w2, b2):
         t.grad = None
 
-    def tanh_block(t):
-        h = F.linear(t, w1, b1)
-        return F.linear(
-            F.gelu(h.float(), approximate="tanh").to(dt), w2, b2
-        )
-
-    yr = tanh_block(x)
+    yr = block(x)  # stock erf path
     yr.float().pow(2).sum().backward()
-    assert float((y.float() - yr.float()).abs().max()) < 0.05
+    assert float((y.float() - yr.float()).abs().max()) < 0.02
     for got, (t, name) in zip(
         grads, [(x, "x"), (w1, "w1"), (b1, "b1"), (w2, "w2"), (b2, "b2")]
     ):
         ref = t.grad.float()
         rel = (got.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
-        assert float(rel) < 5e-2, (name, float(rel))
+        assert float(rel) < 3e-2, (name, float(rel))
+
 
 
 def test_fwd_gemm_vs_reference():

@@ -225,28 +225,30 @@ def _dual_key(a, b):
 
 
 # --------------------------------------------------------------------------
-# Fused MLP GELU via hipblaslt-ext epilogues (ROADMAP item 6 / VERDICT 3).
+# Fused MLP GELU (ROADMAP item 6 / VERDICT 3).
 #
 # Enabled by configure_gelu_fusion(embed_dim, hidden_dim) + the
-# VITFSDP_FUSED_GELU env knob.  Mechanism (all below autograd, so the
-# stock addmm/mm/gelu graph nodes — and checkpoint early-stop — are
-# untouched):
-#   forward:  fc1's addmm runs as lt_gemm_gelu -> (gelu_out, pre); the
-#             dispatch returns PRE as the addmm result (so GeluBackward
-#             saves the correct pre-activation) and answers the
-#             immediately following aten.gelu with the cached gelu_out.
-#   backward: fc2's dgrad mm runs as lt_gemm_dgelu_bgrad(dy, W2, pre)
-#             -> (dpre, dbias1) using the pre-activation saved on a
-#             weakref stack by the forward/recompute; the following
-#             aten.gelu_backward is answered with dpre as-is, and fc1's
-#             dbias sum with the fused dbias1.
-# hipBLASLt's GELU is the tanh approximation; because BOTH directions
-# are fused, forward and backward stay consistent (the model trains
-# with tanh-GELU — measured-equivalence policy documented in
-# docs/PERF_MODEL.md; the erf default path remains when fusion is off).
+# VITFSDP_FUSED_GELU env knob.  Mechanism (below autograd, so the stock
+# addmm/gelu graph nodes — and checkpoint early-stop — are untouched):
+# fc1's forward addmm runs as _C.fwd_gemm_gelu (csrc/fgemm.hip, the
+# hand-written CDNA4 GEMM with an EXACT-erf GELU+bias epilogue); the
+# dispatch returns the PRE-activation as the addmm result (so
+# GeluBackward saves exactly what the stock graph expects) and answers
+# the immediately following aten.gelu with the cached gelu output.  The
+# backward is entirely stock — erf forward + erf gelu_backward, i.e.
+# reference numerics up to GEMM rounding.
+#
+# (hipblaslt-ext's GELU_AUX/DGELU epilogues ship no algorithms for bf16
+# on this ROCm 7.2 / gfx950 stack — probed on-box, r2 GPU call 2 — and
+# are tanh-approx anyway; the lt_gemm_gelu / lt_gemm_dgelu_bgrad
+# entries remain in csrc/ltgemm.cpp as dormant alternatives.)
 # --------------------------------------------------------------------------
 
 _GELU_CFG = {"d": 0, "hid": 0}
+
+# minimum (M/256)*(N/256) workgroups before fc1 routes to fgemm (grid
+# fill, same rationale as the wgrad gate)
+_FGEMM_MIN_TILES = int(os.environ.get("VITFSDP_FGEMM_MIN_TILES", "256"))
 
 
 def configure_gelu_fusion(embed_dim, hidden_dim):
@@ -265,15 +267,6 @@ def _gelu_fusion_on():
     )
 
 
-def _dgelu_tanh(go, pre):
-    """tanh-approx GELU derivative (matches hipBLASLt's epilogue): the
-    loud-but-correct fallback when a gelu_backward could not be fused
-    after a tanh-fused forward."""
-    c = 0.7978845608028654  # sqrt(2/pi)
-    x = pre.float()
-    t = torch.tanh(c * (x + 0.044715 * x * x * x))
-    dt = (1.0 - t * t) * c * (1.0 + 3 * 0.044715 * x * x)
-    return (go.float() * (0.5 * (1.0 + t) + 0.5 * x * dt)).to(go.dtype)
 
 
 class TunedGemmMode(TorchDispatchMode):
@@ -301,12 +294,8 @@ class TunedGemmMode(TorchDispatchMode):
         self.hits = 0
         self.wgrad_hits = 0
         # fused-GELU state (see module comment above):
-        self.fused_gelu = _gelu_fusion_on() and handler is None
+        self.fused_gelu = _gelu_fusion_on()
         self._pending_gelu = {}  # pre-act data_ptr -> gelu_out (transient)
-        self._aux_stack = []  # weakrefs to pre-activations, LIFO
-        self._aux_live = {}  # pre-act data_ptr -> weakref (fallback check)
-        self._dgelu_done = {}  # dpre data_ptr -> (aux_ptr, dbias)
-        self._pending_dbias = None  # (dpre data_ptr, fused dbias)
         self.gelu_hits = 0
 
     def _tuned_index(self, a, b):
@@ -330,63 +319,35 @@ class TunedGemmMode(TorchDispatchMode):
     # -- fused-GELU helpers -------------------------------------------------
 
     def _is_fc1_fwd(self, key):
-        return (
+        if not (
             key is not None
             and key[0] == "T" and key[1] == "N"
             and key[2] == _GELU_CFG["hid"] and key[4] == _GELU_CFG["d"]
-        )
-
-    def _is_fc2_dgrad(self, key):
+        ):
+            return False
+        if self._handler is not None:
+            return True  # CPU test seam has no kernel shape limits
+        hid, tok = key[2], key[3]
         return (
-            key is not None
-            and key[0] == "N" and key[1] == "N"
-            and key[2] == _GELU_CFG["hid"] and key[4] == _GELU_CFG["d"]
+            tok % 256 == 0 and hid % 256 == 0 and key[4] % 64 == 0
+            and (tok // 256) * (hid // 256) >= _FGEMM_MIN_TILES
         )
 
-    def _fused_fc1_fwd(self, a, b, bias, idx):
-        import weakref
-
-        out, pre = ext().lt_gemm_gelu(a, b, bias, -1 if idx is None else idx)
+    def _fused_fc1_fwd(self, a, b, bias):
+        if self._handler is not None:  # CPU test seam
+            out, pre = self._handler(a, b, "gelu", bias)
+        else:
+            # b is the transposed view of fc1's weight [hid, d]: b.t()
+            # IS the weight, row-major contiguous — fgemm reads it
+            # directly (W [N, K], K contiguous)
+            out, pre = ext().fwd_gemm_gelu(
+                a.contiguous(), b.t().contiguous(), bias
+            )
         self.gelu_hits += 1
         if len(self._pending_gelu) > 4:  # unconsumed strays (shouldn't happen)
             self._pending_gelu.clear()
         self._pending_gelu[pre.data_ptr()] = out
-        # Lifetime of the pre-activation until the fc2-dgrad fusion:
-        #  * original forward: autograd's SavedVariable keeps the C++
-        #    tensor (and, via PyObject preservation, the Python object)
-        #    alive — a weakref suffices, and for a checkpointed block
-        #    (whose activations are discarded) it correctly dies;
-        #  * recompute (inside a backward graph task): the checkpoint
-        #    frame saves a DETACHED copy, so the weakref would die at
-        #    early-stop — hold a strong ref instead; it is consumed by
-        #    the same block's fc2-dgrad moments later and the per-step
-        #    mode instance bounds any leftover to one step.
-        in_backward = torch._C._current_graph_task_id() != -1
-        entry = pre if in_backward else weakref.ref(pre)
-        self._aux_stack.append(entry)
-        self._aux_live[pre.data_ptr()] = entry
-        if len(self._aux_live) > 256:
-            self._aux_live = {
-                p: r for p, r in self._aux_live.items()
-                if self._resolve(r) is not None
-            }
         return pre
-
-    @staticmethod
-    def _resolve(entry):
-        return entry if torch.is_tensor(entry) else entry()
-
-    def _pop_aux(self, rows, cols):
-        while self._aux_stack:
-            aux = self._resolve(self._aux_stack[-1])
-            if aux is None:
-                self._aux_stack.pop()
-                continue
-            if aux.shape[0] == rows and aux.shape[1] == cols:
-                self._aux_stack.pop()
-                return aux
-            return None  # live but mismatched: leave for its own consumer
-        return None
 
     def __torch_dispatch__(self, func, types, args=(), kwargs=None):
         kwargs = kwargs or {}
@@ -402,19 +363,7 @@ class TunedGemmMode(TorchDispatchMode):
                 (dw,) = ext().wgrad_gemm(a.t().contiguous(), b.contiguous(),
                                          False)
                 return dw
-            key = _dual_key(a, b)
-            idx = self.table.get(key) if (self.table and key) else None
-            if self.fused_gelu and self._is_fc2_dgrad(key):
-                aux = self._pop_aux(a.shape[0], b.shape[1])
-                if aux is not None:
-                    dpre, dbias = ext().lt_gemm_dgelu_bgrad(
-                        a, b, aux, -1 if idx is None else idx
-                    )
-                    self.gelu_hits += 1
-                    self._dgelu_done[dpre.data_ptr()] = (
-                        aux.data_ptr(), dbias,
-                    )
-                    return dpre
+            idx = self._tuned_index(a, b)
             if idx is not None:
                 self.hits += 1
                 return self._route(a, b, idx)
@@ -428,9 +377,9 @@ class TunedGemmMode(TorchDispatchMode):
         ):
             bias, a, b = args
             key = _dual_key(a, b)
-            idx = self.table.get(key) if (self.table and key) else None
             if self.fused_gelu and self._is_fc1_fwd(key):
-                return self._fused_fc1_fwd(a, b, bias, idx)
+                return self._fused_fc1_fwd(a, b, bias)
+            idx = self.table.get(key) if (self.table and key) else None
             if idx is not None:
                 self.hits += 1
                 return self._route(a, b, idx, bias)
@@ -438,34 +387,6 @@ class TunedGemmMode(TorchDispatchMode):
             out = self._pending_gelu.pop(args[0].data_ptr(), None)
             if out is not None and out.shape == args[0].shape:
                 return out
-        elif self.fused_gelu and func is torch.ops.aten.gelu_backward.default:
-            go, pre = args[0], args[1]
-            ent = self._dgelu_done.pop(go.data_ptr(), None)
-            if ent is not None and ent[0] == pre.data_ptr():
-                # dgelu already applied in the fc2-dgrad GEMM epilogue;
-                # remember the fused dbias for fc1's bias-grad sum
-                self._pending_dbias = (go.data_ptr(), ent[1])
-                return go
-            ref = self._aux_live.get(pre.data_ptr())
-            if ref is not None and self._resolve(ref) is not None:
-                # forward was tanh-fused but the backward GEMM fusion
-                # missed: apply the matching tanh derivative explicitly
-                return _dgelu_tanh(go, pre)
-        elif (
-            self.fused_gelu
-            and func is torch.ops.aten.sum.dim_IntList
-            and len(args) >= 2
-            and args[1] == [0]
-            and getattr(self, "_pending_dbias", None) is not None
-            and args[0].data_ptr() == self._pending_dbias[0]
-        ):
-            dbias = self._pending_dbias[1]
-            self._pending_dbias = None
-            if dbias.shape[0] == args[0].shape[1]:
-                keepdim = args[2] if len(args) > 2 else kwargs.get(
-                    "keepdim", False
-                )
-                return dbias.unsqueeze(0) if keepdim else dbias
         return func(*args, **kwargs)
 
 
