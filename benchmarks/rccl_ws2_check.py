@@ -86,11 +86,73 @@ def run(steps, ws1=False):
         torch.distributed.destroy_process_group()
 
 
+
+
+def loopback():
+    """RCCL cannot place two ranks on one GPU ("Duplicate GPU detected",
+    RCCL 2.26 — gpurun_out/ws2_full.log), so inside a 1-GPU lease this
+    exercises the REAL nccl(=RCCL) backend at ws=1: init, dual
+    communicator creation (CommContext), and the exact collective calls
+    + message sizes of the ViT-10B FSDP step (async
+    all_gather_into_tensor of the 629 MB bf16 unit payload,
+    reduce_scatter_tensor, scalar all-reduce) issued directly on the
+    groups (the engine's own ws=1 path short-circuits them)."""
+    import time
+
+    import torch
+    import torch.distributed as dist
+
+    from vit_10b_fsdp_example_amd import dist as xdist
+    from vit_10b_fsdp_example_amd.parallel import CommContext
+
+    os.environ.setdefault("WORLD_SIZE", "1")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29529")
+    device = xdist.init_distributed()
+    assert dist.is_initialized() and dist.get_backend() == "nccl"
+    CommContext.reset()
+    ctx = CommContext.get()
+    assert ctx.gather_group is not None and ctx.reduce_group is not None
+
+    unit = 314_639_360  # ViT-10B block params
+    shard = torch.randn(unit, device=device).to(torch.bfloat16)
+    full = torch.empty(unit, device=device, dtype=torch.bfloat16)
+    out = {}
+    t0 = time.time()
+    for _ in range(3):
+        w = dist.all_gather_into_tensor(full, shard, group=ctx.gather_group,
+                                        async_op=True)
+        w.wait()
+    torch.cuda.synchronize()
+    out["allgather_ms"] = round((time.time() - t0) / 3 * 1e3, 2)
+    assert torch.equal(full, shard)
+
+    grads = torch.randn(unit, device=device).to(torch.bfloat16)
+    red = torch.empty(unit, device=device, dtype=torch.bfloat16)
+    t0 = time.time()
+    for _ in range(3):
+        w = dist.reduce_scatter_tensor(red, grads, group=ctx.reduce_group,
+                                       async_op=True)
+        w.wait()
+    torch.cuda.synchronize()
+    out["reducescatter_ms"] = round((time.time() - t0) / 3 * 1e3, 2)
+    assert torch.equal(red, grads)
+
+    s = torch.ones((), device=device)
+    dist.all_reduce(s, group=ctx.reduce_group)
+    out["scalar_allreduce"] = float(s)
+    print(json.dumps({"loopback": out, "backend": "nccl(RCCL)"}), flush=True)
+    dist.destroy_process_group()
+
+
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=4)
     ap.add_argument("--ws1", action="store_true",
                     help="single-process reference run")
+    ap.add_argument("--loopback", action="store_true",
+                    help="ws=1 through the real RCCL backend (1-GPU lease)")
     ap.add_argument("--ag", default=None, choices=["allgather", "p2p"])
     ap.add_argument("--rs", default=None, choices=["reducescatter", "p2p"])
     a = ap.parse_args()
@@ -102,4 +164,7 @@ if __name__ == "__main__":
         for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
                   "MASTER_PORT"):
             os.environ.pop(k, None)
-    run(a.steps, ws1=a.ws1)
+    if a.loopback:
+        loopback()
+    else:
+        run(a.steps, ws1=a.ws1)
