@@ -42,6 +42,15 @@ struct FgemmShared {
   short b_tile[2][kBN][kSK];  // W tile, [n][k]
 };
 
+// one b128 LDS read via asm so the consuming MFMAs can be paced with
+// counted lgkmcnt waits (wgemm.hip scheme; a plain dereference lets the
+// compiler hoist every read and then drain with lgkmcnt(0))
+__device__ __forceinline__ bf16x8 lds_read_b128(unsigned byte_addr) {
+  bf16x8 v;
+  asm volatile("ds_read_b128 %0, %1 offset:0" : "=v"(v) : "v"(byte_addr));
+  return v;
+}
+
 // epilogue modes: 0 = plain, 1 = +bias, 2 = +bias + exact-erf GELU
 // (writes the pre-activation to aux for the backward — unlike the
 // hipblaslt-ext GELU_AUX_BIAS epilogue this keeps the reference's erf
@@ -120,34 +129,62 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
   write_tiles(0);
   __syncthreads();
 
+  // byte bases of the LDS tiles for the asm-paced fragment reads
+  const unsigned lds0 = (unsigned)__builtin_amdgcn_groupstaticsize();
+  const unsigned a_base0 = lds0;
+  const unsigned b_base0 = lds0 + (unsigned)sizeof(sm.a_tile);
+  constexpr unsigned kABufBytes = sizeof(sm.a_tile[0]);
+  constexpr unsigned kBBufBytes = sizeof(sm.b_tile[0]);
+
   for (int ks = 0; ks < n_ksteps; ++ks) {
     const int buf = ks & 1;
+    const unsigned a_base = a_base0 + (unsigned)buf * kABufBytes;
+    const unsigned b_base = b_base0 + (unsigned)buf * kBBufBytes;
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
       // next K-step's global loads issue at the start of the second
       // chunk: latency hides under its MFMAs (wgemm.hip T14 scheme)
       if (kc == 1 && ks + 1 < n_ksteps) issue_loads((long)(ks + 1) * kBK);
       const int kcol = kc * 32 + seg * 8;
-      // B fragments for this chunk (4 plain b128 LDS reads, reused
-      // across all 8 m fragments)
+      // B fragments for this chunk (4 b128 reads, reused across all 8
+      // m fragments), then the A fragments stream through a 2-deep
+      // ring with counted lgkmcnt waits so MFMAs start as soon as
+      // their own fragment retires (wgemm.hip pacing)
       bf16x8 bf[4];
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
-        bf[ni] = *reinterpret_cast<const bf16x8*>(
-            &sm.b_tile[buf][wn * 64 + ni * 16 + col][kcol]);
+        bf[ni] = lds_read_b128(
+            b_base + 2u * ((unsigned)((wn * 64 + ni * 16 + col) * kSK + kcol)));
       }
-      __builtin_amdgcn_s_setprio(1);
+      bf16x8 af[2];
+      af[0] = lds_read_b128(
+          a_base + 2u * ((unsigned)((wm * 128 + col) * kSK + kcol)));
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi) {
-        const bf16x8 af = *reinterpret_cast<const bf16x8*>(
-            &sm.a_tile[buf][wm * 128 + mi * 16 + col][kcol]);
+        const int cur = mi & 1;
+        if (mi < 7) {
+          af[cur ^ 1] = lds_read_b128(
+              a_base +
+              2u * ((unsigned)((wm * 128 + (mi + 1) * 16 + col) * kSK + kcol)));
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(af[cur]), "+v"(bf[0]), "+v"(bf[1]), "+v"(bf[2]),
+                         "+v"(bf[3])
+                       : [cnt] "i"(1)
+                       : "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
+                       : "+v"(af[cur])
+                       : [cnt] "i"(0)
+                       : "memory");
+        }
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af, bf[ni], acc[mi][ni], 0, 0, 0);
+              af[cur], bf[ni], acc[mi][ni], 0, 0, 0);
         }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     if (ks + 1 < n_ksteps) write_tiles(buf ^ 1);
     __syncthreads();
