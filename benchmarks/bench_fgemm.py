@@ -34,9 +34,18 @@ def bench(M, N, K, iters=12):
         torch.nn.functional.linear(xx, ww, bb)
     torch.cuda.synchronize(); t_lib = (time.time() - t0) / iters
 
+    # second library baseline through our own hipblaslt-ext call (the
+    # torch pick occasionally lands on a slow algo for some shapes)
+    torch.cuda.synchronize(); t0 = time.time()
+    for i in range(iters):
+        xx, ww, bb = sets[i % 4]
+        C.lt_gemm(xx, ww.t(), -1, bb)
+    torch.cuda.synchronize(); t_lt = (time.time() - t0) / iters
+
     fl = 2.0 * K * M * N
     print(f"M{M} N{N} K{K}: ours {t_ours*1e3:7.2f} ms {fl/t_ours/1e12:7.0f} TF"
-          f" | hipBLASLt {t_lib*1e3:7.2f} ms {fl/t_lib/1e12:7.0f} TF"
+          f" | torch {t_lib*1e3:7.2f} ms {fl/t_lib/1e12:7.0f} TF"
+          f" | lt-heur {t_lt*1e3:7.2f} ms {fl/t_lt/1e12:7.0f} TF"
           f" | rel_err {rel:.4f}", flush=True)
 
 
