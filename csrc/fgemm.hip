@@ -19,6 +19,8 @@
 #ifndef VITFSDP_KERNELS_ONLY
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
+
+#include <cstdlib>
 #endif
 
 #include "common.h"
@@ -30,16 +32,19 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr int kBM = 256;
 constexpr int kBN = 256;
-constexpr int kBK = 64;
 constexpr int kThreads = 512;  // 8 waves: 2 (m) x 4 (n)
-// LDS tiles are [row][k] with the k row padded +8 elements (16 B): the
-// row stride is then 144 B, so the 16 lanes of a fragment-read group
-// (consecutive rows, same k column) touch distinct bank lines
-constexpr int kSK = kBK + 8;
 
+// BK is a template knob: 64 = fewer barriers, 147 KB LDS, 1 WG/CU;
+// 32 = half the LDS (82 KB) so TWO workgroups co-reside per CU and one
+// WG computes through the other's staging barrier.
+// LDS tiles are [row][k] padded +8 elements: the row stride in dwords
+// is 4*odd, so the 16 lanes of a b128 fragment-read group (consecutive
+// rows, same k column) land on 16 distinct 4-dword bank slots.
+template <int BK>
 struct FgemmShared {
-  short a_tile[2][kBM][kSK];  // X tile, [m][k]
-  short b_tile[2][kBN][kSK];  // W tile, [n][k]
+  static constexpr int SK = BK + 8;
+  short a_tile[2][kBM][SK];  // X tile, [m][k]
+  short b_tile[2][kBN][SK];  // W tile, [n][k]
 };
 
 // one b128 LDS read via asm so the consuming MFMAs can be paced with
@@ -56,8 +61,8 @@ __device__ __forceinline__ bf16x8 lds_read_b128(unsigned byte_addr) {
 // hipblaslt-ext GELU_AUX_BIAS epilogue this keeps the reference's erf
 // GELU numerics; ~10 extra VALU ops/element disappear under the MFMA
 // phases of a compute-bound tile)
-template <int EPI>
-__global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
+template <int EPI, int BK>
+__global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
     const short* __restrict__ a,     // [M, K]
     const short* __restrict__ b,     // [N, K]
     const short* __restrict__ bias,  // optional [N], may be null
@@ -65,8 +70,10 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
     short* __restrict__ aux,         // EPI==2: pre-GELU [M, N]
     int M, int N, int K) {
   constexpr bool WITH_BIAS = EPI >= 1;
+  constexpr int kSK = FgemmShared<BK>::SK;
+  constexpr int kChunks = BK / 32;
   HIP_DYNAMIC_SHARED(char, smem_raw)
-  FgemmShared& sm = *reinterpret_cast<FgemmShared*>(smem_raw);
+  FgemmShared<BK>& sm = *reinterpret_cast<FgemmShared<BK>*>(smem_raw);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -95,14 +102,15 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
 
   // staging: thread t loads rows (t>>3) + i*64, k-cols (t&7)*8 (+16 B
   // vectors).  256 rows x 64 k per tile, 512 threads -> 4 vectors each.
-  constexpr int kVecs = kBM * kBK / 8 / kThreads;  // 4
-  const int s_r0 = tid >> 3;
-  const int s_c8 = (tid & 7) * 8;
+  constexpr int kVecs = kBM * BK / 8 / kThreads;  // 4 at BK=64, 2 at BK=32
+  constexpr int kPerRow = BK / 8;  // 16B vectors per tile row
+  const int s_r0 = tid / kPerRow;
+  const int s_c8 = (tid % kPerRow) * 8;
   bf16x8 a_st[kVecs], b_st[kVecs];
   auto issue_loads = [&](long k_base) {
 #pragma unroll
     for (int i = 0; i < kVecs; ++i) {
-      const long row = s_r0 + i * (kThreads / 8);
+      const long row = s_r0 + i * (kThreads / kPerRow);
       a_st[i] =
           *reinterpret_cast<const bf16x8*>(&a[(m0 + row) * K + k_base + s_c8]);
       b_st[i] =
@@ -112,7 +120,7 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
   auto write_tiles = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < kVecs; ++i) {
-      const int row = s_r0 + i * (kThreads / 8);
+      const int row = s_r0 + i * (kThreads / kPerRow);
       *reinterpret_cast<bf16x8*>(&sm.a_tile[buf][row][s_c8]) = a_st[i];
       *reinterpret_cast<bf16x8*>(&sm.b_tile[buf][row][s_c8]) = b_st[i];
     }
@@ -124,7 +132,7 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int n_ksteps = K / kBK;
+  const int n_ksteps = K / BK;
   issue_loads(0);
   write_tiles(0);
   __syncthreads();
@@ -141,10 +149,11 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
     const unsigned a_base = a_base0 + (unsigned)buf * kABufBytes;
     const unsigned b_base = b_base0 + (unsigned)buf * kBBufBytes;
 #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      // next K-step's global loads issue at the start of the second
+    for (int kc = 0; kc < kChunks; ++kc) {
+      // next K-step's global loads issue at the start of the last
       // chunk: latency hides under its MFMAs (wgemm.hip T14 scheme)
-      if (kc == 1 && ks + 1 < n_ksteps) issue_loads((long)(ks + 1) * kBK);
+      if (kc == kChunks - 1 && ks + 1 < n_ksteps)
+        issue_loads((long)(ks + 1) * BK);
       const int kcol = kc * 32 + seg * 8;
       // B fragments for this chunk (4 b128 reads, reused across all 8
       // m fragments), then the A fragments stream through a 2-deep
@@ -221,20 +230,62 @@ __global__ __launch_bounds__(kThreads, 1) void fgemm_abt_kernel(
 }
 
 #ifdef VITFSDP_KERNELS_ONLY
-template __global__ void fgemm_abt_kernel<0>(const short*, const short*,
-                                             const short*, short*, short*,
-                                             int, int, int);
-template __global__ void fgemm_abt_kernel<1>(const short*, const short*,
-                                             const short*, short*, short*,
-                                             int, int, int);
-template __global__ void fgemm_abt_kernel<2>(const short*, const short*,
-                                             const short*, short*, short*,
-                                             int, int, int);
+template __global__ void fgemm_abt_kernel<0, 64>(const short*, const short*,
+                                                 const short*, short*, short*,
+                                                 int, int, int);
+template __global__ void fgemm_abt_kernel<1, 64>(const short*, const short*,
+                                                 const short*, short*, short*,
+                                                 int, int, int);
+template __global__ void fgemm_abt_kernel<2, 64>(const short*, const short*,
+                                                 const short*, short*, short*,
+                                                 int, int, int);
+template __global__ void fgemm_abt_kernel<0, 32>(const short*, const short*,
+                                                 const short*, short*, short*,
+                                                 int, int, int);
 #endif
 
 }  // namespace
 
 #ifndef VITFSDP_KERNELS_ONLY
+namespace {
+
+// runtime BK selection (A/B knob): 64 (default) or 32 (2 WG/CU)
+int fgemm_bk() {
+  static int bk = [] {
+    const char* e = std::getenv("VITFSDP_FGEMM_BK");
+    return (e != nullptr && std::atoi(e) == 32) ? 32 : 64;
+  }();
+  return bk;
+}
+
+template <int EPI, int BK>
+void launch_fgemm(const short* x, const short* w, const short* bias, short* c,
+                  short* aux, long M, long N, long K, hipStream_t stream) {
+  static bool attr_set = [] {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&fgemm_abt_kernel<EPI, BK>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, sizeof(FgemmShared<BK>));
+    return true;
+  }();
+  (void)attr_set;
+  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
+  hipLaunchKernelGGL((fgemm_abt_kernel<EPI, BK>), grid, dim3(kThreads),
+                     sizeof(FgemmShared<BK>), stream, x, w, bias, c, aux,
+                     (int)M, (int)N, (int)K);
+}
+
+template <int EPI>
+void launch_fgemm_bk(const short* x, const short* w, const short* bias,
+                     short* c, short* aux, long M, long N, long K,
+                     hipStream_t stream) {
+  if (fgemm_bk() == 32 && K % 32 == 0)
+    launch_fgemm<EPI, 32>(x, w, bias, c, aux, M, N, K, stream);
+  else
+    launch_fgemm<EPI, 64>(x, w, bias, c, aux, M, N, K, stream);
+}
+
+}  // namespace
+
 torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
                        c10::optional<torch::Tensor> bias) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous(),
@@ -244,7 +295,7 @@ torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
               "fwd_gemm: x[M,K], w[N,K] expected");
   const long M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % kBK == 0,
+  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % 64 == 0,
               "fwd_gemm: needs M%256==0, N%256==0, K%64==0 (got ", M, ",", N,
               ",", K, ")");
   auto c = torch::empty({M, N}, x.options());
@@ -256,28 +307,15 @@ torch::Tensor fwd_gemm(torch::Tensor x, torch::Tensor w,
                 bias_c.numel() == N, "fwd_gemm: bias must be bf16 [N]");
     bias_ptr = (const short*)bias_c.data_ptr();
   }
-  static bool attr_set = [] {
-    for (auto fn : {reinterpret_cast<const void*>(&fgemm_abt_kernel<0>),
-                    reinterpret_cast<const void*>(&fgemm_abt_kernel<1>),
-                    reinterpret_cast<const void*>(&fgemm_abt_kernel<2>)}) {
-      (void)hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
-                                sizeof(FgemmShared));
-    }
-    return true;
-  }();
-  (void)attr_set;
-  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
   auto stream = at::cuda::getCurrentCUDAStream();
   if (bias_ptr) {
-    hipLaunchKernelGGL(fgemm_abt_kernel<1>, grid, dim3(kThreads),
-                       sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), bias_ptr,
-                       (short*)c.data_ptr(), nullptr, (int)M, (int)N, (int)K);
+    launch_fgemm_bk<1>((const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       bias_ptr, (short*)c.data_ptr(), nullptr, M, N, K,
+                       stream);
   } else {
-    hipLaunchKernelGGL(fgemm_abt_kernel<0>, grid, dim3(kThreads),
-                       sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), nullptr,
-                       (short*)c.data_ptr(), nullptr, (int)M, (int)N, (int)K);
+    launch_fgemm_bk<0>((const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                       nullptr, (short*)c.data_ptr(), nullptr, M, N, K,
+                       stream);
   }
   HIP_CHECK_LAST();
   return c;
@@ -297,17 +335,14 @@ std::vector<torch::Tensor> fwd_gemm_gelu(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1) &&
               bias.numel() == w.size(0), "fwd_gemm_gelu: x[M,K], w[N,K]");
   const long M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % kBK == 0,
+  TORCH_CHECK(M % kBM == 0 && N % kBN == 0 && K % 64 == 0,
               "fwd_gemm_gelu: needs M%256==0, N%256==0, K%64==0");
   auto c = torch::empty({M, N}, x.options());
   auto aux = torch::empty({M, N}, x.options());
-  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(fgemm_abt_kernel<2>, grid, dim3(kThreads),
-                     sizeof(FgemmShared), stream, (const short*)x.data_ptr(),
-                     (const short*)w.data_ptr(), (const short*)bias.data_ptr(),
-                     (short*)c.data_ptr(), (short*)aux.data_ptr(), (int)M,
-                     (int)N, (int)K);
+  launch_fgemm_bk<2>((const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     (const short*)bias.data_ptr(), (short*)c.data_ptr(),
+                     (short*)aux.data_ptr(), M, N, K, stream);
   HIP_CHECK_LAST();
   return {c, aux};
 }

@@ -128,19 +128,19 @@ int main() {
   short* c = alloc_fill<short>((size_t)M * N);
   short* aux = alloc_fill<short>((size_t)M * N);
   short* bias = alloc_fill<short>((size_t)N);
-  for (auto fn : {reinterpret_cast<const void*>(&fgemm_abt_kernel<0>),
-                  reinterpret_cast<const void*>(&fgemm_abt_kernel<2>)}) {
+  for (auto fn : {reinterpret_cast<const void*>(&fgemm_abt_kernel<0, 64>),
+                  reinterpret_cast<const void*>(&fgemm_abt_kernel<2, 64>)}) {
     CHECK(hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
-                              sizeof(FgemmShared)));
+                              sizeof(FgemmShared<64>)));
   }
   dim3 grid((unsigned)(M / 256), (unsigned)(N / 256));
-  hipLaunchKernelGGL(fgemm_abt_kernel<0>, grid, dim3(512),
-                     sizeof(FgemmShared), 0, x, w, nullptr, c, nullptr, M, N,
+  hipLaunchKernelGGL((fgemm_abt_kernel<0, 64>), grid, dim3(512),
+                     sizeof(FgemmShared<64>), 0, x, w, nullptr, c, nullptr, M, N,
                      K);
   CHECK(hipGetLastError());
   CHECK(hipDeviceSynchronize());
-  hipLaunchKernelGGL(fgemm_abt_kernel<2>, grid, dim3(512),
-                     sizeof(FgemmShared), 0, x, w, bias, c, aux, M, N, K);
+  hipLaunchKernelGGL((fgemm_abt_kernel<2, 64>), grid, dim3(512),
+                     sizeof(FgemmShared<64>), 0, x, w, bias, c, aux, M, N, K);
   CHECK(hipGetLastError());
   CHECK(hipDeviceSynchronize());
   printf("fgemm ok\n");
