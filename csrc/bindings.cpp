@@ -42,16 +42,19 @@ torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logits,
                                 torch::Tensor target, torch::Tensor lse);
 
 std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, double scale);
+                                    torch::Tensor v, double scale,
+                                    double p_drop, long seed);
 std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    double scale);
+                                    double scale, double p_drop, long seed);
 std::vector<torch::Tensor> fmha_fwd_qkv(torch::Tensor qkv, long num_heads,
-                                        double scale);
+                                        double scale, double p_drop,
+                                        long seed);
 torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
                            torch::Tensor o, torch::Tensor lse,
-                           long num_heads, double scale);
+                           long num_heads, double scale, double p_drop,
+                           long seed);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
@@ -93,13 +96,27 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place scale by device scalar");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused softmax CE forward");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused softmax CE backward");
-  m.def("fmha_fwd", &fmha_fwd, "flash attention forward (bf16, head_dim<=192)");
-  m.def("fmha_bwd", &fmha_bwd, "flash attention backward (fused FA2-style)");
+  m.def("fmha_fwd", &fmha_fwd,
+        "flash attention forward (bf16, head_dim<=192; in-kernel "
+        "attention dropout via a counter-hash mask)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
+  m.def("fmha_bwd", &fmha_bwd,
+        "flash attention backward (fused FA2-style; regenerates the "
+        "dropout mask from the seed)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
   m.def("fmha_fwd_qkv", &fmha_fwd_qkv,
         "flash attention forward on the fused [B,T,3,H,D] qkv projection "
-        "(zero-copy strided IO, O returned as [B,T,E])");
+        "(zero-copy strided IO, O returned as [B,T,E])",
+        py::arg("qkv"), py::arg("num_heads"), py::arg("scale"),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
   m.def("fmha_bwd_qkv", &fmha_bwd_qkv,
-        "flash attention backward producing the fused dqkv [B,T,3,H,D]");
+        "flash attention backward producing the fused dqkv [B,T,3,H,D]",
+        py::arg("dout"), py::arg("qkv"), py::arg("o"), py::arg("lse"),
+        py::arg("num_heads"), py::arg("scale"), py::arg("p_drop") = 0.0,
+        py::arg("seed") = 0);
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("wgrad_gemm", &wgrad_gemm,

@@ -110,13 +110,30 @@ struct QkvStrides {
   long ob, oh, ot;  // o (and dO) strides
 };
 
-template <int D>
+// Attention-dropout mask: stateless integer hash of (seed, global q row
+// incl. bh, k column) so forward and both backward kernels regenerate
+// the identical mask with no stored bits.  keep iff hash >= threshold
+// (threshold = p * 2^32); integer-exact and reproduced bit-for-bit by
+// the python reference in tests/test_gpu_kernels.py.
+__device__ __forceinline__ unsigned dropout_hash(unsigned seed, unsigned qg,
+                                                 unsigned kg) {
+  unsigned x = seed ^ (qg * 0x9E3779B9u) ^ (kg * 0x85EBCA6Bu);
+  x ^= x >> 16;
+  x *= 0x7FEB352Du;
+  x ^= x >> 15;
+  x *= 0x846CA68Bu;
+  x ^= x >> 16;
+  return x;
+}
+
+template <int D, bool DROP = false>
 // min 3 waves/SIMD: the double-buffered staging otherwise lands at
 // 182 regs -> 2 waves; forcing 3 costs a few prologue spills at most
 __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse_out, int T, int H, QkvStrides st, float scale) {
+    float* __restrict__ lse_out, int T, int H, QkvStrides st, float scale,
+    unsigned drop_thresh = 0, float drop_rscale = 1.f, unsigned seed = 0) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQSub = S::QSub;
@@ -316,6 +333,23 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
       row_sum += __shfl_xor(row_sum, 32);
       l_run[qs] = l_run[qs] * alpha + row_sum;
 
+      if (DROP) {
+        // dropout on P (post-softmax, reference nn.Dropout semantics):
+        // the row sum above stays UNMASKED (it is the softmax
+        // normalizer); the PV path gets the masked, 1/(1-p)-scaled P
+        const unsigned qg = (unsigned)(bh * T + q_row0 + 16 * qs + col);
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const unsigned kg = (unsigned)(k_base + 16 * kk + seg * 4 + r);
+            p[kk][r] = dropout_hash(seed, qg, kg) >= drop_thresh
+                           ? p[kk][r] * drop_rscale
+                           : 0.f;
+          }
+        }
+      }
+
       // pack P to bf16 pairs (consecutive k): pk[kk][rr] holds
       // (k = 16*kk + seg*4 + 2*rr, +1) for q = col
       unsigned pk[2][2];
@@ -452,12 +486,13 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_fwd_kernel(
   }
 }
 
-template <int D>
+template <int D, bool DROP = false>
 __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, int T, int H, QkvStrides st, float scale) {
+    short* __restrict__ dq, int T, int H, QkvStrides st, float scale,
+    unsigned drop_thresh = 0, float drop_rscale = 1.f, unsigned seed = 0) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQSub = S::QSub;
@@ -592,10 +627,19 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
           for (int h = 0; h < 2; ++h) {
             const int r = 2 * rr + h;
             float ds = 0.f;
-            if (k_base + 16 * kk + seg * 4 + r < T) {
+            const int k_idx = k_base + 16 * kk + seg * 4 + r;
+            if (k_idx < T) {
               const float pv =
                   __expf(st_frag[kk][r] * scale - lse_q[qs]);
-              ds = scale * pv * (dpt_frag[kk][r] - delta_q[qs]);
+              float dp = dpt_frag[kk][r];
+              if (DROP) {
+                // dP = dPD * D (gradient through the dropout mask)
+                const unsigned qg = (unsigned)(bh * T + q_row0 + 16 * qs + col);
+                dp = dropout_hash(seed, qg, (unsigned)k_idx) >= drop_thresh
+                         ? dp * drop_rscale
+                         : 0.f;
+              }
+              ds = scale * pv * (dp - delta_q[qs]);
             }
             packed |= (unsigned)f32_to_bf16(ds) << (16 * h);
           }
@@ -694,13 +738,14 @@ __global__ __launch_bounds__(kBlockThreads, 3) void fmha_bwd_dq_kernel(
   }
 }
 
-template <int D>
+template <int D, bool DROP = false>
 __global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int T, int H,
-    QkvStrides st, float scale) {
+    QkvStrides st, float scale,
+    unsigned drop_thresh = 0, float drop_rscale = 1.f, unsigned seed = 0) {
   using S = FmhaShapes<D>;
   constexpr int DP = S::DP, NKC = S::NKC, NC = S::NC;
   constexpr int kQStream = 32;  // q rows streamed per iteration
@@ -824,7 +869,18 @@ __global__ __launch_bounds__(kBlockThreads, 2) void fmha_bwd_dkv_kernel(
         float pt = 0.f, dst = 0.f;
         if (q_ok) {
           pt = __expf(st_frag[qq][r] * scale - l);
-          dst = scale * pt * (dpt_frag[qq][r] - dlt);
+          float dp = dpt_frag[qq][r];
+          if (DROP) {
+            const unsigned qg = (unsigned)(bh * T + q_idx);
+            const unsigned kg = (unsigned)(k_row0 + seg * 4 + r);
+            const bool keep = dropout_hash(seed, qg, kg) >= drop_thresh;
+            dp = keep ? dp * drop_rscale : 0.f;
+            dst = scale * pt * (dp - dlt);
+            // dV uses the MASKED P (the forward multiplied PD into V)
+            pt = keep ? pt * drop_rscale : 0.f;
+          } else {
+            dst = scale * pt * (dp - dlt);
+          }
         }
         sm.pt_tile[wave][seg * 4 + r][qq * 16 + col] =
             (short)f32_to_bf16(pt);
@@ -1025,22 +1081,23 @@ __global__ __launch_bounds__(64) void mfma_probe_kernel(
 
 #ifdef VITFSDP_KERNELS_ONLY
 // explicit instantiations so -Rpass-analysis reports the real configs
-template __global__ void fmha_fwd_kernel<160>(const short*, const short*,
-                                              const short*, short*, float*,
-                                              int, int, QkvStrides, float);
-template __global__ void fmha_fwd_kernel<64>(const short*, const short*,
-                                             const short*, short*, float*,
-                                             int, int, QkvStrides, float);
-template __global__ void fmha_bwd_dq_kernel<160>(const short*, const short*,
-                                                 const short*, const short*,
-                                                 const float*, const float*,
-                                                 short*, int, int, QkvStrides,
-                                                 float);
-template __global__ void fmha_bwd_dkv_kernel<160>(const short*, const short*,
-                                                  const short*, const short*,
-                                                  const float*, const float*,
-                                                  short*, short*, int, int,
-                                                  QkvStrides, float);
+template __global__ void fmha_fwd_kernel<160, false>(
+    const short*, const short*, const short*, short*, float*, int, int,
+    QkvStrides, float, unsigned, float, unsigned);
+template __global__ void fmha_fwd_kernel<160, true>(
+    const short*, const short*, const short*, short*, float*, int, int,
+    QkvStrides, float, unsigned, float, unsigned);
+template __global__ void fmha_fwd_kernel<64, false>(
+    const short*, const short*, const short*, short*, float*, int, int,
+    QkvStrides, float, unsigned, float, unsigned);
+template __global__ void fmha_bwd_dq_kernel<160, false>(
+    const short*, const short*, const short*, const short*, const float*,
+    const float*, short*, int, int, QkvStrides, float, unsigned, float,
+    unsigned);
+template __global__ void fmha_bwd_dkv_kernel<160, false>(
+    const short*, const short*, const short*, const short*, const float*,
+    const float*, short*, short*, int, int, QkvStrides, float, unsigned,
+    float, unsigned);
 #else
 struct FmhaArgs {
   const short *q, *k, *v;
@@ -1048,16 +1105,33 @@ struct FmhaArgs {
   int B, H, T;
   QkvStrides st;
   float scale;
+  float p_drop = 0.f;   // attention dropout probability
+  unsigned seed = 0;    // dropout mask seed
 };
+
+inline unsigned drop_threshold(float p) {
+  if (p <= 0.f) return 0u;
+  double t = (double)p * 4294967296.0;
+  return t >= 4294967295.0 ? 4294967295u : (unsigned)t;
+}
 
 template <int D>
 void launch_fmha_fwd(const FmhaArgs& a, torch::Tensor& lse) {
   dim3 grid((unsigned)((long)a.B * a.H),
             (a.T + FmhaShapes<D>::QTile - 1) / FmhaShapes<D>::QTile);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(fmha_fwd_kernel<D>, grid, dim3(kBlockThreads),
-                     sizeof(typename FmhaShapes<D>::Shared), stream, a.q, a.k,
-                     a.v, a.o, lse.data_ptr<float>(), a.T, a.H, a.st, a.scale);
+  if (a.p_drop > 0.f) {
+    hipLaunchKernelGGL((fmha_fwd_kernel<D, true>), grid, dim3(kBlockThreads),
+                       sizeof(typename FmhaShapes<D>::Shared), stream, a.q,
+                       a.k, a.v, a.o, lse.data_ptr<float>(), a.T, a.H, a.st,
+                       a.scale, drop_threshold(a.p_drop),
+                       1.f / (1.f - a.p_drop), a.seed);
+  } else {
+    hipLaunchKernelGGL((fmha_fwd_kernel<D, false>), grid, dim3(kBlockThreads),
+                       sizeof(typename FmhaShapes<D>::Shared), stream, a.q,
+                       a.k, a.v, a.o, lse.data_ptr<float>(), a.T, a.H, a.st,
+                       a.scale, 0u, 1.f, 0u);
+  }
   HIP_CHECK_LAST();
 }
 
@@ -1068,18 +1142,37 @@ void launch_fmha_bwd(const FmhaArgs& a, const torch::Tensor& lse,
   using S = FmhaShapes<D>;
   auto stream = at::cuda::getCurrentCUDAStream();
   const unsigned BH = (unsigned)((long)a.B * a.H);
+  const unsigned thr = drop_threshold(a.p_drop);
+  const float rs = a.p_drop > 0.f ? 1.f / (1.f - a.p_drop) : 1.f;
   dim3 grid_dq(BH, (a.T + S::QTile - 1) / S::QTile);
-  hipLaunchKernelGGL(fmha_bwd_dq_kernel<D>, grid_dq, dim3(kBlockThreads),
-                     sizeof(typename S::SharedDQ), stream, a.q, a.k, a.v,
-                     a.o /*dO*/, lse.data_ptr<float>(),
-                     delta.data_ptr<float>(), dq, a.T, a.H, a.st, a.scale);
+  if (a.p_drop > 0.f) {
+    hipLaunchKernelGGL((fmha_bwd_dq_kernel<D, true>), grid_dq,
+                       dim3(kBlockThreads), sizeof(typename S::SharedDQ),
+                       stream, a.q, a.k, a.v, a.o /*dO*/,
+                       lse.data_ptr<float>(), delta.data_ptr<float>(), dq,
+                       a.T, a.H, a.st, a.scale, thr, rs, a.seed);
+  } else {
+    hipLaunchKernelGGL((fmha_bwd_dq_kernel<D, false>), grid_dq,
+                       dim3(kBlockThreads), sizeof(typename S::SharedDQ),
+                       stream, a.q, a.k, a.v, a.o /*dO*/,
+                       lse.data_ptr<float>(), delta.data_ptr<float>(), dq,
+                       a.T, a.H, a.st, a.scale, 0u, 1.f, 0u);
+  }
   HIP_CHECK_LAST();
   dim3 grid_dkv(BH, (a.T + 63) / 64);
-  hipLaunchKernelGGL(fmha_bwd_dkv_kernel<D>, grid_dkv, dim3(kBlockThreads),
-                     sizeof(typename S::SharedDKV), stream, a.q, a.k, a.v,
-                     a.o /*dO*/, lse.data_ptr<float>(),
-                     delta.data_ptr<float>(), dk, dv, a.T, a.H, a.st,
-                     a.scale);
+  if (a.p_drop > 0.f) {
+    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<D, true>), grid_dkv,
+                       dim3(kBlockThreads), sizeof(typename S::SharedDKV),
+                       stream, a.q, a.k, a.v, a.o /*dO*/,
+                       lse.data_ptr<float>(), delta.data_ptr<float>(), dk, dv,
+                       a.T, a.H, a.st, a.scale, thr, rs, a.seed);
+  } else {
+    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<D, false>), grid_dkv,
+                       dim3(kBlockThreads), sizeof(typename S::SharedDKV),
+                       stream, a.q, a.k, a.v, a.o /*dO*/,
+                       lse.data_ptr<float>(), delta.data_ptr<float>(), dk, dv,
+                       a.T, a.H, a.st, a.scale, 0u, 1.f, 0u);
+  }
   HIP_CHECK_LAST();
 }
 
@@ -1135,7 +1228,8 @@ torch::Tensor run_rowdot(const torch::Tensor& dout, const torch::Tensor& o,
 // ---- [B,H,T,D] contiguous API (kernel unit tests, generic use) ----
 
 std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, double scale) {
+                                    torch::Tensor v, double scale,
+                                    double p_drop, long seed) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
               v.is_contiguous());
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "fmha_fwd: bf16 only");
@@ -1148,7 +1242,8 @@ std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
   auto lse = torch::empty({B, H, T}, q.options().dtype(torch::kFloat32));
   FmhaArgs a{(const short*)q.data_ptr(), (const short*)k.data_ptr(),
              (const short*)v.data_ptr(), (short*)o.data_ptr(),
-             B, H, T, contiguous_strides(H, T, D), (float)scale};
+             B, H, T, contiguous_strides(H, T, D), (float)scale,
+             (float)p_drop, (unsigned)seed};
   VITFSDP_FMHA_DISPATCH(D, launch_fmha_fwd<kD>(a, lse));
   return {o, lse};
 }
@@ -1156,7 +1251,7 @@ std::vector<torch::Tensor> fmha_fwd(torch::Tensor q, torch::Tensor k,
 std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    double scale) {
+                                    double scale, double p_drop, long seed) {
   TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
   const int B = q.size(0), H = q.size(1), T = q.size(2), D = q.size(3);
   const QkvStrides st = contiguous_strides(H, T, D);
@@ -1193,7 +1288,7 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
   auto dv = torch::empty_like(v);
   FmhaArgs a{(const short*)q.data_ptr(), (const short*)k.data_ptr(),
              (const short*)v.data_ptr(), (short*)dout.data_ptr(),
-             B, H, T, st, (float)scale};
+             B, H, T, st, (float)scale, (float)p_drop, (unsigned)seed};
   VITFSDP_FMHA_DISPATCH(
       D, launch_fmha_bwd<kD>(a, lse, delta, (short*)dq.data_ptr(),
                              (short*)dk.data_ptr(), (short*)dv.data_ptr()));
@@ -1205,7 +1300,8 @@ std::vector<torch::Tensor> fmha_bwd(torch::Tensor dout, torch::Tensor q,
 // o comes back [B, T, H, D] (== [B, T, E]); dO arrives the same way.
 
 std::vector<torch::Tensor> fmha_fwd_qkv(torch::Tensor qkv, long num_heads,
-                                        double scale) {
+                                        double scale, double p_drop,
+                                        long seed) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 5 &&
                   qkv.size(2) == 3 && qkv.size(3) == num_heads,
               "fmha_fwd_qkv: expected contiguous [B, T, 3, H, D]");
@@ -1226,14 +1322,16 @@ std::vector<torch::Tensor> fmha_fwd_qkv(torch::Tensor qkv, long num_heads,
   auto lse = torch::empty({B, H, T}, qkv.options().dtype(torch::kFloat32));
   const short* base = (const short*)qkv.data_ptr();
   FmhaArgs a{base, base + (long)H * D, base + 2L * H * D,
-             (short*)o.data_ptr(), B, H, T, st, (float)scale};
+             (short*)o.data_ptr(), B, H, T, st, (float)scale,
+             (float)p_drop, (unsigned)seed};
   VITFSDP_FMHA_DISPATCH(D, launch_fmha_fwd<kD>(a, lse));
   return {o, lse};
 }
 
 torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
                            torch::Tensor o, torch::Tensor lse,
-                           long num_heads, double scale) {
+                           long num_heads, double scale, double p_drop,
+                           long seed) {
   TORCH_CHECK(dout.is_cuda() && dout.is_contiguous() && qkv.is_contiguous());
   const int B = qkv.size(0), T = qkv.size(1), H = qkv.size(3),
             D = qkv.size(4);
@@ -1250,7 +1348,8 @@ torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
   const short* base = (const short*)qkv.data_ptr();
   short* dbase = (short*)dqkv.data_ptr();
   FmhaArgs a{base, base + (long)H * D, base + 2L * H * D,
-             (short*)dout.data_ptr(), B, H, T, st, (float)scale};
+             (short*)dout.data_ptr(), B, H, T, st, (float)scale,
+             (float)p_drop, (unsigned)seed};
   VITFSDP_FMHA_DISPATCH(
       D, launch_fmha_bwd<kD>(a, lse, delta, dbase, dbase + (long)H * D,
                              dbase + 2L * H * D));

@@ -605,3 +605,87 @@ def test_fwd_gemm_gelu_erf_vs_reference():
     out_ref = torch.nn.functional.gelu(pre_ref)  # erf-exact
     assert float((pre.float() - pre_ref).abs().max()) < 0.05
     assert float((out.float() - out_ref).abs().max()) < 0.05
+
+
+def test_fmha_dropout_in_kernel_fwd_bwd():
+    """In-kernel attention dropout: the kernel must match the math
+    composition evaluated with the EXACT mask (bit-reproduced from the
+    counter hash), forward and backward."""
+    from vit_10b_fsdp_example_amd.ops.attention import (
+        dropout_mask_reference,
+    )
+
+    torch.manual_seed(0)
+    B, H, T, D = 2, 3, 128, 160
+    p, seed, scale = 0.3, 98765, D ** -0.5
+    mk = lambda: (torch.randn(B, H, T, D, device=_dev()) * 0.5).to(
+        torch.bfloat16
+    )
+    q, k, v = mk(), mk(), mk()
+    for t in (q, k, v):
+        t.requires_grad_(True)
+
+    o, lse = EXT.fmha_fwd(q, k, v, scale, p, seed)
+
+    # exact-mask reference in fp32
+    mask = torch.empty(B, H, T, T, device=_dev())
+    for b in range(B):
+        for h in range(H):
+            rows = torch.arange(T) + (b * H + h) * T
+            mask[b, h] = dropout_mask_reference(
+                seed, rows, torch.arange(T), p
+            ).float().to(_dev())
+    qf, kf, vf = (t.detach().float().requires_grad_(True) for t in (q, k, v))
+    probs = torch.softmax(qf @ kf.transpose(-2, -1) * scale, dim=-1)
+    o_ref = (probs * mask / (1 - p)) @ vf
+    err = (o.float() - o_ref).abs().max()
+    assert float(err) < 0.05, float(err)
+
+    # same seed -> identical output; different seed -> different
+    o2, _ = EXT.fmha_fwd(q, k, v, scale, p, seed)
+    assert torch.equal(o, o2)
+    o3, _ = EXT.fmha_fwd(q, k, v, scale, p, seed + 1)
+    assert not torch.equal(o, o3)
+
+    # measured drop rate ~ p (the output of a row with all-ones V would
+    # be the masked row-mean; check via probs-mask stats instead)
+    keep_frac = mask.mean().item()
+    assert abs(keep_frac - (1 - p)) < 0.01
+
+    # backward vs the exact-mask reference
+    do = torch.randn_like(o)
+    dq, dk, dv = EXT.fmha_bwd(do, q, k, v, o, lse, scale, p, seed)
+    o_ref.backward(do.float())
+    for got, ref, name in [(dq, qf.grad, "dq"), (dk, kf.grad, "dk"),
+                           (dv, vf.grad, "dv")]:
+        rel = (got.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
+        assert float(rel) < 6e-2, (name, float(rel))
+
+
+def test_attention_dropout_autograd_path():
+    """attention_qkv with dropout_p > 0 must use the flash kernel (no
+    O(T^2) fallback) and train: grads finite, loss decreasing on a tiny
+    overfit."""
+    from vit_10b_fsdp_example_amd.ops import attention_qkv
+
+    torch.manual_seed(1)
+    B, T, H, D = 2, 64, 2, 64
+    qkv = (torch.randn(B, T, 3, H, D, device=_dev()) * 0.2).to(
+        torch.bfloat16
+    ).requires_grad_(True)
+    torch.manual_seed(7)
+    out = attention_qkv(qkv, H, dropout_p=0.25, training=True)
+    assert out.shape == (B, T, H * D)
+    out.float().pow(2).sum().backward()
+    assert torch.isfinite(qkv.grad.float()).all()
+    # reproducibility through torch.manual_seed (seed drawn from CPU RNG)
+    g1 = qkv.grad.clone()
+    qkv.grad = None
+    torch.manual_seed(7)
+    out2 = attention_qkv(qkv, H, dropout_p=0.25, training=True)
+    out2.float().pow(2).sum().backward()
+    assert torch.equal(out, out2) and torch.equal(g1, qkv.grad)
+    # eval mode: no dropout -> deterministic equality with p=0 path
+    oe = attention_qkv(qkv, H, dropout_p=0.25, training=False)
+    o0 = attention_qkv(qkv, H, dropout_p=0.0, training=False)
+    assert torch.equal(oe, o0)

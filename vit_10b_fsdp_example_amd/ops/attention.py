@@ -7,35 +7,44 @@ softmax(Q K^T * d^-0.5) V with optional attention dropout
 own CDNA4 flash-style kernel (MFMA 16x16x32 bf16 tiles, LDS-staged K/V,
 online softmax, O(T) memory; backward recomputes the forward tiles).
 
+Attention dropout runs IN-KERNEL (round 2): the mask is a stateless
+integer hash of (seed, global q row, k column) applied to the
+post-softmax P on the PV path, regenerated identically by the backward
+kernels — no mask storage, O(T) memory preserved
+(csrc/fmha.hip dropout_hash; the seed is drawn from torch's CPU RNG so
+torch.manual_seed reproduces runs).
+
 No Triton, no aotriton SDPA: the CPU/no-ext fallback is an explicit
 math composition (matmul + softmax), which is also the numerics
 reference for the kernel tests.
 """
 
-import warnings
-
 import torch
 
 from ._extension import ext, use_hip
 
-_DROPOUT_WARNED = False
+
+def dropout_mask_reference(seed, bh_q, k, p):
+    """Bit-exact python reproduction of csrc/fmha.hip dropout_hash for
+    the GPU tests: returns the keep mask (bool) for global row indices
+    ``bh_q`` [rows] x key columns ``k`` [cols] at probability p."""
+    M = 0xFFFFFFFF
+    qg = bh_q.to(torch.int64).reshape(-1, 1)
+    kg = k.to(torch.int64).reshape(1, -1)
+    x = (int(seed) ^ ((qg * 0x9E3779B9) & M) ^ ((kg * 0x85EBCA6B) & M)) & M
+    x = x ^ (x >> 16)
+    x = (x * 0x7FEB352D) & M
+    x = x ^ (x >> 15)
+    x = (x * 0x846CA68B) & M
+    x = x ^ (x >> 16)
+    thresh = min(int(p * 4294967296.0), 4294967295)
+    return x >= thresh
 
 
-def _warn_dropout_fallback():
-    """--att_dropout > 0 routes to the O(T^2) math composition (the
-    flash kernel has no in-kernel RNG yet).  The 10B training recipe
-    uses att_dropout 0.0 (reference run_vit_training.py:346 default),
-    so this is off the measured path — but it must never be a silent
-    30x attention slowdown + O(T^2) memory change."""
-    global _DROPOUT_WARNED
-    if not _DROPOUT_WARNED:
-        _DROPOUT_WARNED = True
-        warnings.warn(
-            "attention dropout > 0: falling back from the flash kernel "
-            "to the explicit-math attention path (O(T^2) memory, "
-            "slower). The reference recipe uses --att_dropout 0.",
-            stacklevel=3,
-        )
+def _draw_seed():
+    # CPU RNG so torch.manual_seed makes dropout reproducible, without
+    # a device sync
+    return int(torch.randint(0, 2**31 - 1, (1,)).item())
 
 
 def math_attention(q, k, v, scale=None, dropout_p=0.0, training=False):
@@ -53,41 +62,42 @@ def math_attention(q, k, v, scale=None, dropout_p=0.0, training=False):
 
 class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
-        o, lse = ext().fmha_fwd(q, k, v, scale)
+    def forward(ctx, q, k, v, scale, p_drop, seed):
+        o, lse = ext().fmha_fwd(q, k, v, scale, p_drop, seed)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
+        ctx.p_drop = p_drop
+        ctx.seed = seed
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = ext().fmha_bwd(do.contiguous(), q, k, v, o, lse, ctx.scale)
-        return dq, dk, dv, None
+        dq, dk, dv = ext().fmha_bwd(
+            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.p_drop, ctx.seed
+        )
+        return dq, dk, dv, None, None, None
 
 
 def attention(q, k, v, scale=None, dropout_p=0.0, training=False):
     """Attention core on [B, H, T, D] tensors.
 
-    GPU: our flash-style HIP kernel (dropout_p must be 0 there for now —
-    the 10B recipe uses att_dropout 0.0; nonzero dropout falls back to
-    the math path with a warning-free explicit mask).
+    GPU: our flash-style HIP kernel, attention dropout in-kernel.
     CPU: math composition.
     """
     if scale is None:
         scale = q.shape[-1] ** -0.5
     use_kernel = (
-        dropout_p == 0.0
-        and q.dtype in (torch.bfloat16, torch.float16)
+        q.dtype in (torch.bfloat16, torch.float16)
         and use_hip(q, k, v)
         and hasattr(ext(), "fmha_fwd")
     )
     if use_kernel:
+        p = dropout_p if training else 0.0
+        seed = _draw_seed() if p > 0.0 else 0
         return _FlashAttentionFn.apply(
-            q.contiguous(), k.contiguous(), v.contiguous(), scale
+            q.contiguous(), k.contiguous(), v.contiguous(), scale, p, seed
         )
-    if dropout_p > 0.0 and training and use_hip(q):
-        _warn_dropout_fallback()
     return math_attention(q, k, v, scale, dropout_p, training)
 
 
@@ -98,43 +108,48 @@ class _FlashAttentionQkvFn(torch.autograd.Function):
     copies on either side of the attention core."""
 
     @staticmethod
-    def forward(ctx, qkv, num_heads, scale):
-        o, lse = ext().fmha_fwd_qkv(qkv, num_heads, scale)
+    def forward(ctx, qkv, num_heads, scale, p_drop, seed):
+        o, lse = ext().fmha_fwd_qkv(qkv, num_heads, scale, p_drop, seed)
         ctx.save_for_backward(qkv, o, lse)
         ctx.num_heads = num_heads
         ctx.scale = scale
+        ctx.p_drop = p_drop
+        ctx.seed = seed
         return o
 
     @staticmethod
     def backward(ctx, do):
         qkv, o, lse = ctx.saved_tensors
         dqkv = ext().fmha_bwd_qkv(
-            do.contiguous(), qkv, o, lse, ctx.num_heads, ctx.scale
+            do.contiguous(), qkv, o, lse, ctx.num_heads, ctx.scale,
+            ctx.p_drop, ctx.seed,
         )
-        return dqkv, None, None
+        return dqkv, None, None, None, None
 
 
 def attention_qkv(qkv, num_heads, scale=None, dropout_p=0.0, training=False):
     """Attention on the fused qkv projection.
 
     qkv: [B, T, 3, H, D] (a free reshape view of the qkv Linear output);
-    returns [B, T, H*D].  GPU bf16 uses the strided flash kernels; the
-    fallback path permutes to [B, H, T, D] and runs the math composition.
+    returns [B, T, H*D].  GPU bf16 uses the strided flash kernels with
+    in-kernel attention dropout; the fallback path permutes to
+    [B, H, T, D] and runs the math composition.
     """
     B, T, three, H, D = qkv.shape
     assert three == 3 and H == num_heads
     if scale is None:
         scale = D ** -0.5
     use_kernel = (
-        dropout_p == 0.0
-        and qkv.dtype in (torch.bfloat16,)
+        qkv.dtype in (torch.bfloat16,)
         and use_hip(qkv)
         and hasattr(ext(), "fmha_fwd_qkv")
     )
     if use_kernel:
-        return _FlashAttentionQkvFn.apply(qkv.contiguous(), num_heads, scale)
-    if dropout_p > 0.0 and training and use_hip(qkv):
-        _warn_dropout_fallback()
+        p = dropout_p if training else 0.0
+        seed = _draw_seed() if p > 0.0 else 0
+        return _FlashAttentionQkvFn.apply(
+            qkv.contiguous(), num_heads, scale, p, seed
+        )
     q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)
     o = math_attention(q, k, v, scale, dropout_p, training)
     return o.transpose(1, 2).reshape(B, T, H * D)
