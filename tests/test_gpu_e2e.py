@@ -273,3 +273,31 @@ def test_ws1_alias_and_deferred_clip_step():
         losses.append(float(loss.detach()))
     assert all(l == l for l in losses), f"NaN loss: {losses}"
     assert losses[-1] < losses[0] - 0.3, f"loss not decreasing: {losses}"
+
+
+def test_rccl_loopback_subprocess():
+    """The real nccl(=RCCL) backend at ws=1: init, dual communicator
+    creation, async all_gather_into_tensor / reduce_scatter_tensor at
+    the ViT-10B unit payload (629 MB bf16).  Subprocess so the process
+    group cannot leak into other tests.  (True multi-rank RCCL needs
+    >1 GPU: RCCL 2.26 rejects duplicate devices —
+    profiles/raw/rccl_ws2_duplicate_gpu.log.)"""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    env["MASTER_PORT"] = "29541"
+    res = subprocess.run(
+        [sys.executable, "benchmarks/rccl_ws2_check.py", "--loopback"],
+        capture_output=True, text=True, cwd=repo, timeout=420, env=env,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    line = [l for l in res.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["backend"] == "nccl(RCCL)"
+    assert out["loopback"]["scalar_allreduce"] == 1.0
+    assert out["loopback"]["allgather_ms"] > 0
