@@ -94,18 +94,37 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
   const int wm = wave >> 2;   // 0..1: wave's 128-row m strip
   const int wn = wave & 3;    // 0..3: wave's 64-col n strip
 
-  // XCD-aware tile swizzle (guide T1, same as wgemm.hip): give each XCD
-  // a contiguous run of tiles so its co-resident blocks share operand
-  // panels in its private L2
+  // XCD-aware 2D tile clustering: the dispatcher places block bid on
+  // XCD bid%8, so each XCD's co-resident ~32 workgroups are
+  // cid = bid>>3 consecutive.  Give each XCD a contiguous 2D
+  // sub-rectangle of the tile grid and walk it in GROUP_M-wide bands:
+  // the resident set then covers a GM x (32/GM) RECTANGLE, reusing A
+  // panels GM-fold and B panels (32/GM)-fold in the XCD's private L2.
+  // (The r1 M-major run reused only B; at 1 workgroup/CU the A streams
+  // alone exceed the per-CU HBM budget — MfmaUtil measured 44%.)
   int bx = blockIdx.x, by = blockIdx.y;
   {
-    const int nwg = gridDim.x * gridDim.y;
-    if ((nwg & 7) == 0) {
-      const int bid = (int)(blockIdx.x + blockIdx.y * gridDim.x);
+    const int gx = gridDim.x, gy = gridDim.y;
+    const int nwg = gx * gy;
+    const int bid = (int)(blockIdx.x + blockIdx.y * gx);
+    int sx = 2, sy = 4;  // 8 XCD sub-rectangles
+    if (gy % 4 != 0) {
+      if (gy % 2 == 0) { sx = 4; sy = 2; }
+      else { sx = 8; sy = 1; }
+    }
+    if ((nwg & 7) == 0 && gx % sx == 0 && gy % sy == 0) {
+      const int xcd = bid & 7, cid = bid >> 3;
+      const int lw = gx / sx, lh = gy / sy;
+      const int gm = (lw % 4 == 0) ? 4 : ((lw % 2 == 0) ? 2 : 1);
+      const int per_band = gm * lh;
+      const int band = cid / per_band, r = cid % per_band;
+      bx = (xcd % sx) * lw + band * gm + (r % gm);
+      by = (xcd / sx) * lh + r / gm;
+    } else if ((nwg & 7) == 0) {
       const int cpx = nwg >> 3;
       const int swz = (bid & 7) * cpx + (bid >> 3);
-      bx = swz % gridDim.x;
-      by = swz / gridDim.x;
+      bx = swz % gx;
+      by = swz / gx;
     }
   }
   const long m0 = (long)bx * kBM;
