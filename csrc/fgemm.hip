@@ -42,7 +42,10 @@ constexpr int kThreads = 512;  // 8 waves: 2 (m) x 4 (n)
 // rows, same k column) land on 16 distinct 4-dword bank slots.
 template <int BK>
 struct FgemmShared {
-  static constexpr int SK = BK + 8;
+  // BK=64 stages via DirectToLds (global_load_lds 16B): lane-packed
+  // rows force the PACKED stride, with bank spread done by an in-row
+  // granule XOR instead of padding.  BK=32 keeps the padded layout.
+  static constexpr int SK = BK == 64 ? 64 : BK + 8;
   short a_tile[2][kBM][SK];  // X tile, [m][k]
   short b_tile[2][kBN][SK];  // W tile, [n][k]
 };
@@ -55,6 +58,11 @@ struct FgemmShared {
 // upper 8 rows makes the two halves bank-disjoint.
 template <int BK>
 __device__ __forceinline__ int swz(int row, int col) {
+  if (BK == 64) {
+    // full granule spread: the 16 rows of a b128 fragment group land
+    // on each bank quad exactly twice (the 2-cycle optimum for 256 B)
+    return col ^ ((row & 7) * 8);
+  }
   return col ^ (((row >> 3) & 1) * (BK / 2));
 }
 
@@ -130,16 +138,44 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
   const long m0 = (long)bx * kBM;
   const long n0 = (long)by * kBN;
 
-  // staging: thread t loads rows (t>>3) + i*64, k-cols (t&7)*8 (+16 B
-  // vectors).  256 rows x 64 k per tile, 512 threads -> 4 vectors each.
+  // staging:
+  //  * BK=64: DirectToLds — each global_load_lds(16B) call deposits the
+  //    wave's 64 lanes contiguously (1 KB = 8 packed rows), so lane l
+  //    covers (row = R + (l>>3), granule l&7) and the global address is
+  //    permuted per-lane to realize the bank-spread XOR.  No staging
+  //    registers, no separate LDS-write phase; synced by vmcnt+barrier.
+  //  * BK=32: classic register staging (padded layout).
   constexpr int kVecs = kBM * BK / 8 / kThreads;  // 4 at BK=64, 2 at BK=32
   constexpr int kPerRow = BK / 8;  // 16B vectors per tile row
   const int s_r0 = tid / kPerRow;
   const int s_c8 = (tid % kPerRow) * 8;
-  bf16x8 a_st[kVecs], b_st[kVecs];
-  auto issue_loads = [&](long k_base) {
+  constexpr int kRowsPerCall = 1024 / (BK * 2);  // 8 at BK=64
+  constexpr int kCalls = kBM / (8 * kRowsPerCall);  // per wave: 4 at BK=64
+  bf16x8 a_st[BK == 64 ? 1 : kVecs], b_st[BK == 64 ? 1 : kVecs];
+  auto issue_dtl = [&](int buf, long k_base) {
+    const int l = lane;
+    const int rsub = l >> 3;
 #pragma unroll
-    for (int i = 0; i < kVecs; ++i) {
+    for (int i = 0; i < kCalls; ++i) {
+      const int R = (wave * kCalls + i) * kRowsPerCall;
+      const int row = R + rsub;
+      const int g = (l & 7) ^ (row & 7);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              &a[(m0 + row) * K + k_base + g * 8]),
+          (__attribute__((address_space(3))) void*)&sm.a_tile[buf][R][0], 16,
+          0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              &b[(n0 + row) * K + k_base + g * 8]),
+          (__attribute__((address_space(3))) void*)&sm.b_tile[buf][R][0], 16,
+          0, 0);
+    }
+  };
+  auto issue_loads = [&](long k_base) {
+    if (BK == 64) return;  // unused (DirectToLds path)
+#pragma unroll
+    for (int i = 0; i < kVecs && BK != 64; ++i) {
       const long row = s_r0 + i * (kThreads / kPerRow);
       a_st[i] =
           *reinterpret_cast<const bf16x8*>(&a[(m0 + row) * K + k_base + s_c8]);
@@ -148,8 +184,9 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
     }
   };
   auto write_tiles = [&](int buf) {
+    if (BK == 64) return;
 #pragma unroll
-    for (int i = 0; i < kVecs; ++i) {
+    for (int i = 0; i < kVecs && BK != 64; ++i) {
       const int row = s_r0 + i * (kThreads / kPerRow);
       const int cc = swz<BK>(row, s_c8);
       *reinterpret_cast<bf16x8*>(&sm.a_tile[buf][row][cc]) = a_st[i];
@@ -164,8 +201,13 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
     for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int n_ksteps = K / BK;
-  issue_loads(0);
-  write_tiles(0);
+  if (BK == 64) {
+    issue_dtl(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    issue_loads(0);
+    write_tiles(0);
+  }
   __syncthreads();
 
   // byte bases of the LDS tiles for the asm-paced fragment reads
@@ -181,10 +223,13 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
     const unsigned b_base = b_base0 + (unsigned)buf * kBBufBytes;
 #pragma unroll
     for (int kc = 0; kc < kChunks; ++kc) {
-      // next K-step's global loads issue at the start of the last
-      // chunk: latency hides under its MFMAs (wgemm.hip T14 scheme)
-      if (kc == kChunks - 1 && ks + 1 < n_ksteps)
+      // next K-step's loads issue early: DirectToLds at the FIRST
+      // chunk (no registers held), register staging at the last
+      if (BK == 64) {
+        if (kc == 0 && ks + 1 < n_ksteps) issue_dtl(buf ^ 1, (long)(ks + 1) * BK);
+      } else if (kc == kChunks - 1 && ks + 1 < n_ksteps) {
         issue_loads((long)(ks + 1) * BK);
+      }
       const int kcol = kc * 32 + seg * 8;
       // B fragments for this chunk (4 b128 reads, reused across all 8
       // m fragments), then the A fragments stream through a 2-deep
@@ -230,7 +275,12 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
     }
-    if (ks + 1 < n_ksteps) write_tiles(buf ^ 1);
+    if (BK == 64) {
+      if (ks + 1 < n_ksteps)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    } else if (ks + 1 < n_ksteps) {
+      write_tiles(buf ^ 1);
+    }
     __syncthreads();
   }
 
