@@ -56,6 +56,7 @@ torch::Tensor fmha_bwd_qkv(torch::Tensor dout, torch::Tensor qkv,
                            long num_heads, double scale, double p_drop,
                            long seed);
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b);
+torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> wgrad_gemm(torch::Tensor a, torch::Tensor b,
                                       bool with_bias);
@@ -118,6 +119,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("num_heads"), py::arg("scale"), py::arg("p_drop") = 0.0,
         py::arg("seed") = 0);
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("mfma32_probe", &mfma32_probe, "32x32x16 bf16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("wgrad_gemm", &wgrad_gemm,
         "C = A^T B weight-gradient GEMM (bf16, tr16 transpose reads)");

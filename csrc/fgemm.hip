@@ -314,6 +314,155 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
   }
 }
 
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// Wide-MFMA variant: mfma_f32_32x32x16_bf16 halves the MFMA instruction
+// count per FLOP (the 16x16 kernel measured VALUBusy 27% competing with
+// MFMA issue at MfmaUtil 53%).  BK=64 + DirectToLds staging + the same
+// clustered mapping; fragment layout pinned by mfma32_probe
+// (tests/test_gpu_kernels.py::test_mfma32_probe_layout).
+template <int EPI>
+__global__ __launch_bounds__(kThreads, 1) void fgemm_abt_mi32_kernel(
+    const short* __restrict__ a, const short* __restrict__ b,
+    const short* __restrict__ bias, short* __restrict__ c,
+    short* __restrict__ aux, int M, int N, int K) {
+  constexpr bool WITH_BIAS = EPI >= 1;
+  constexpr int BK = 64;
+  HIP_DYNAMIC_SHARED(char, smem_raw)
+  FgemmShared<BK>& sm = *reinterpret_cast<FgemmShared<BK>*>(smem_raw);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31;  // fragment row (A) / column (B)
+  const int kseg = lane >> 5;  // k-slice 8*kseg within a 16-k chunk
+  const int wm = wave >> 2;   // 0..1: wave's 128-row m strip
+  const int wn = wave & 3;    // 0..3: wave's 64-col n strip
+
+  int bx = blockIdx.x, by = blockIdx.y;
+  {
+    const int gx = gridDim.x, gy = gridDim.y;
+    const int nwg = gx * gy;
+    const int bid = (int)(blockIdx.x + blockIdx.y * gx);
+    int sx = 2, sy = 4;
+    if (gy % 4 != 0) {
+      if (gy % 2 == 0) { sx = 4; sy = 2; }
+      else { sx = 8; sy = 1; }
+    }
+    if ((nwg & 7) == 0 && gx % sx == 0 && gy % sy == 0) {
+      const int xcd = bid & 7, cid = bid >> 3;
+      const int lw = gx / sx, lh = gy / sy;
+      const int gm = (lw % 4 == 0) ? 4 : ((lw % 2 == 0) ? 2 : 1);
+      const int per_band = gm * lh;
+      const int band = cid / per_band, r = cid % per_band;
+      bx = (xcd % sx) * lw + band * gm + (r % gm);
+      by = (xcd / sx) * lh + r / gm;
+    } else if ((nwg & 7) == 0) {
+      const int cpx = nwg >> 3;
+      const int swzb = (bid & 7) * cpx + (bid >> 3);
+      bx = swzb % gx;
+      by = swzb / gx;
+    }
+  }
+  const long m0 = (long)bx * kBM;
+  const long n0 = (long)by * kBN;
+
+  // DirectToLds staging (same scheme as the 16x16 kernel)
+  auto issue_dtl = [&](int buf, long k_base) {
+    const int rsub = lane >> 3;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int R = (wave * 4 + i) * 8;
+      const int row = R + rsub;
+      const int g = (lane & 7) ^ (row & 7);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              &a[(m0 + row) * K + k_base + g * 8]),
+          (__attribute__((address_space(3))) void*)&sm.a_tile[buf][R][0], 16,
+          0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              &b[(n0 + row) * K + k_base + g * 8]),
+          (__attribute__((address_space(3))) void*)&sm.b_tile[buf][R][0], 16,
+          0, 0);
+    }
+  };
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = f32x16{};
+
+  const int n_ksteps = K / BK;
+  issue_dtl(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int ks = 0; ks < n_ksteps; ++ks) {
+    const int buf = ks & 1;
+    if (ks + 1 < n_ksteps) issue_dtl(buf ^ 1, (long)(ks + 1) * BK);
+#pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {  // 4 chunks of k=16
+      const int kcol = kc * 16 + kseg * 8;
+      bf16x8 bf[2];
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int br = wn * 64 + ni * 32 + l31;
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            &sm.b_tile[buf][br][swz<BK>(br, kcol)]);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int ar = wm * 128 + mi * 32 + l31;
+        const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+            &sm.a_tile[buf][ar][swz<BK>(ar, kcol)]);
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, bf[ni], acc[mi][ni], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (ks + 1 < n_ksteps)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  // epilogue: lane holds col l31 of rows (lane>>5)*4 + g*8 + r
+  float bias_v[2];
+  if (WITH_BIAS) {
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      bias_v[ni] = bf16_to_f32(
+          (unsigned short)bias[n0 + wn * 64 + ni * 32 + l31]);
+    }
+  }
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m0 + wm * 128 + mi * 32 + kseg * 4 + g * 8 + r;
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const long n = n0 + wn * 64 + ni * 32 + l31;
+          float v = acc[mi][ni][g * 4 + r];
+          if (WITH_BIAS) v += bias_v[ni];
+          if (EPI == 2) {
+            aux[m * N + n] = (short)f32_to_bf16(v);
+            v = 0.5f * v * (1.f + erff(v * 0.70710678118654752f));
+          }
+          c[m * N + n] = (short)f32_to_bf16(v);
+        }
+      }
+    }
+  }
+}
+
 #ifdef VITFSDP_KERNELS_ONLY
 template __global__ void fgemm_abt_kernel<0, 64>(const short*, const short*,
                                                  const short*, short*, short*,
@@ -327,6 +476,9 @@ template __global__ void fgemm_abt_kernel<2, 64>(const short*, const short*,
 template __global__ void fgemm_abt_kernel<0, 32>(const short*, const short*,
                                                  const short*, short*, short*,
                                                  int, int, int);
+template __global__ void fgemm_abt_mi32_kernel<0>(const short*, const short*,
+                                                  const short*, short*,
+                                                  short*, int, int, int);
 #endif
 
 }  // namespace
@@ -341,6 +493,16 @@ int fgemm_bk() {
     return (e != nullptr && std::atoi(e) == 32) ? 32 : 64;
   }();
   return bk;
+}
+
+// MFMA shape selection: 32 (mfma_32x32x16, default — half the MFMA
+// instruction count) or 16 (mfma_16x16x32) via VITFSDP_FGEMM_MI=16
+int fgemm_mi() {
+  static int mi = [] {
+    const char* e = std::getenv("VITFSDP_FGEMM_MI");
+    return (e != nullptr && std::atoi(e) == 16) ? 16 : 32;
+  }();
+  return mi;
 }
 
 template <int EPI, int BK>
@@ -360,11 +522,30 @@ void launch_fgemm(const short* x, const short* w, const short* bias, short* c,
 }
 
 template <int EPI>
+void launch_fgemm_mi32(const short* x, const short* w, const short* bias,
+                       short* c, short* aux, long M, long N, long K,
+                       hipStream_t stream) {
+  static bool attr_set = [] {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&fgemm_abt_mi32_kernel<EPI>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, sizeof(FgemmShared<64>));
+    return true;
+  }();
+  (void)attr_set;
+  dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
+  hipLaunchKernelGGL((fgemm_abt_mi32_kernel<EPI>), grid, dim3(kThreads),
+                     sizeof(FgemmShared<64>), stream, x, w, bias, c, aux,
+                     (int)M, (int)N, (int)K);
+}
+
+template <int EPI>
 void launch_fgemm_bk(const short* x, const short* w, const short* bias,
                      short* c, short* aux, long M, long N, long K,
                      hipStream_t stream) {
   if (fgemm_bk() == 32 && K % 32 == 0)
     launch_fgemm<EPI, 32>(x, w, bias, c, aux, M, N, K, stream);
+  else if (fgemm_mi() == 32)
+    launch_fgemm_mi32<EPI>(x, w, bias, c, aux, M, N, K, stream);
   else
     launch_fgemm<EPI, 64>(x, w, bias, c, aux, M, N, K, stream);
 }

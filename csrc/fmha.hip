@@ -1059,6 +1059,29 @@ __global__ __launch_bounds__(64) void tr16_probe_kernel(
   out[lane * 4 + 3] = (unsigned short)(v[1] >> 16);
 }
 
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// Layout probe for the WIDE MFMA (mfma_f32_32x32x16_bf16): one wave,
+// row-major fp32 A[32][16] / B[16][32], using the assumed gfx9-family
+// fragment maps (A: lane l row l&31, k (l>>5)*8+j; C: lane l col l&31,
+// row (l>>5)*4 + g*8 + r at acc[g*4+r]).  The GPU test compares against
+// torch.matmul to pin the layout before csrc/fgemm.hip relies on it.
+__global__ __launch_bounds__(64) void mfma32_probe_kernel(
+    const float* __restrict__ a, const float* __restrict__ b,
+    float* __restrict__ c) {
+  const int l = threadIdx.x;
+  union { unsigned short u[8]; bf16x8 v; } av, bv;
+  for (int j = 0; j < 8; ++j) {
+    av.u[j] = f32_to_bf16(a[(l & 31) * 16 + (l >> 5) * 8 + j]);
+    bv.u[j] = f32_to_bf16(b[((l >> 5) * 8 + j) * 32 + (l & 31)]);
+  }
+  f32x16 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av.v, bv.v, acc, 0, 0, 0);
+  for (int g = 0; g < 4; ++g)
+    for (int r = 0; r < 4; ++r)
+      c[((l >> 5) * 4 + g * 8 + r) * 32 + (l & 31)] = acc[g * 4 + r];
+}
+
 // Layout probe: one wave computes a single 16x16x32 MFMA from row-major
 // fp32 A[16][32], B[32][16] using the documented fragment maps; the GPU
 // test compares against torch.matmul to pin the layout assumptions.
@@ -1366,6 +1389,22 @@ torch::Tensor tr16_probe(long mode) {
                      (unsigned short*)out.data_ptr(), (int)mode);
   HIP_CHECK_LAST();
   return out;
+}
+
+torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && a.dim() == 2 &&
+              a.size(0) == 32 && a.size(1) == 16);
+  TORCH_CHECK(b.is_cuda() && b.is_contiguous() && b.dim() == 2 &&
+              b.size(0) == 16 && b.size(1) == 32);
+  TORCH_CHECK(a.scalar_type() == torch::kFloat32 &&
+              b.scalar_type() == torch::kFloat32);
+  auto c = torch::zeros({32, 32}, a.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(mfma32_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     a.data_ptr<float>(), b.data_ptr<float>(),
+                     c.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return c;
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {

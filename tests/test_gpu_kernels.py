@@ -689,3 +689,16 @@ def test_attention_dropout_autograd_path():
     oe = attention_qkv(qkv, H, dropout_p=0.25, training=False)
     o0 = attention_qkv(qkv, H, dropout_p=0.0, training=False)
     assert torch.equal(oe, o0)
+
+
+def test_mfma32_probe_layout():
+    """Pin the 32x32x16 bf16 MFMA fragment layout (csrc/fgemm.hip's
+    wide-MFMA variant relies on it)."""
+    torch.manual_seed(0)
+    a = torch.randn(32, 16, device=_dev())
+    b = (torch.arange(16 * 32, device=_dev(), dtype=torch.float32)
+         .reshape(16, 32) % 7) - 3.0 + 0.1 * torch.randn(16, 32, device=_dev())
+    c = EXT.mfma32_probe(a, b)
+    ref = a.to(torch.bfloat16).float() @ b.to(torch.bfloat16).float()
+    err = (c - ref).abs().max() / ref.abs().max()
+    assert float(err) < 2e-2, float(err)
