@@ -495,12 +495,15 @@ int fgemm_bk() {
   return bk;
 }
 
-// MFMA shape selection: 32 (mfma_32x32x16, default — half the MFMA
-// instruction count) or 16 (mfma_16x16x32) via VITFSDP_FGEMM_MI=16
+// MFMA shape selection via VITFSDP_FGEMM_MI: 16 (mfma_16x16x32,
+// default — measured 1109-1168 TF/s) or 32 (mfma_32x32x16 — halves the
+// MFMA instruction count but measured 1048-1082: the longer-latency
+// wide MFMA with only 2 independent n-chains per m-tile loses more to
+// dependency stalls than it gains in issue slots)
 int fgemm_mi() {
   static int mi = [] {
     const char* e = std::getenv("VITFSDP_FGEMM_MI");
-    return (e != nullptr && std::atoi(e) == 16) ? 16 : 32;
+    return (e != nullptr && std::atoi(e) == 32) ? 32 : 16;
   }();
   return mi;
 }
