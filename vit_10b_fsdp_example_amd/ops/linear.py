@@ -377,7 +377,12 @@ class TunedGemmMode(TorchDispatchMode):
         ):
             bias, a, b = args
             key = _dual_key(a, b)
-            if self.fused_gelu and self._is_fc1_fwd(key):
+            if (
+                self.fused_gelu
+                and self._is_fc1_fwd(key)
+                and (self._handler is not None
+                     or hasattr(ext(), "fwd_gemm_gelu"))
+            ):
                 return self._fused_fc1_fwd(a, b, bias)
             idx = self.table.get(key) if (self.table and key) else None
             if idx is not None:
