@@ -349,8 +349,14 @@ def test_att_dropout_with_checkpointing():
     l1, g1 = one_step_grads()
     l2, g2 = one_step_grads()
     assert l1 == l2, f"dropout fwd not deterministic: {l1} vs {l2}"
+    # a WRONG recompute mask flips ~20% of P entries and moves gradients
+    # by O(1); library split-K GEMMs are only ULP-reproducible across
+    # invocations, so pin the seed path with a tight tolerance instead
+    # of bitwise equality
     for a, b in zip(g1, g2):
-        assert torch.equal(a, b), "dropout bwd not deterministic"
+        scale = a.float().abs().max() + 1e-6
+        rel = (a.float() - b.float()).abs().max() / scale
+        assert float(rel) < 1e-3, f"dropout bwd mask mismatch: rel {rel}"
 
     # training health over full steps (clip + AdamW included)
     model, x, y = build()
