@@ -242,38 +242,33 @@ __global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
         bf[ni] = lds_read_b128(
             b_base + 2u * ((unsigned)(br * kSK + swz<BK>(br, kcol))));
       }
-      bf16x8 af[2];
-      {
-        const int ar = wm * 128 + col;
-        af[0] = lds_read_b128(
-            a_base + 2u * ((unsigned)(ar * kSK + swz<BK>(ar, kcol))));
-      }
+      // all 8 A fragments issued up-front: one deep DS pipeline, a
+      // single counted wait per mi instead of a 2-deep ring (the ring
+      // measured no better than plain; this removes the per-mi
+      // round-trips entirely at +32 VGPRs)
+      bf16x8 af[8];
 #pragma unroll
       for (int mi = 0; mi < 8; ++mi) {
-        const int cur = mi & 1;
-        if (mi < 7) {
-          const int ar = wm * 128 + (mi + 1) * 16 + col;
-          af[cur ^ 1] = lds_read_b128(
-              a_base + 2u * ((unsigned)(ar * kSK + swz<BK>(ar, kcol))));
-          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
-                       : "+v"(af[cur]), "+v"(bf[0]), "+v"(bf[1]), "+v"(bf[2]),
-                         "+v"(bf[3])
-                       : [cnt] "i"(1)
-                       : "memory");
-        } else {
-          asm volatile("s_waitcnt lgkmcnt(%[cnt])"
-                       : "+v"(af[cur])
-                       : [cnt] "i"(0)
-                       : "memory");
-        }
-        __builtin_amdgcn_s_setprio(1);
+        const int ar = wm * 128 + mi * 16 + col;
+        af[mi] = lds_read_b128(
+            a_base + 2u * ((unsigned)(ar * kSK + swz<BK>(ar, kcol))));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)"
+                   : "+v"(af[0]), "+v"(af[1]), "+v"(af[2]), "+v"(af[3]),
+                     "+v"(af[4]), "+v"(af[5]), "+v"(af[6]), "+v"(af[7]),
+                     "+v"(bf[0]), "+v"(bf[1]), "+v"(bf[2]), "+v"(bf[3])
+                   :
+                   : "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi) {
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[cur], bf[ni], acc[mi][ni], 0, 0, 0);
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     if (BK == 64) {
       if (ks + 1 < n_ksteps)
