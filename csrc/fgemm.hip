@@ -40,18 +40,14 @@ constexpr int kThreads = 512;  // 8 waves: 2 (m) x 4 (n)
 // LDS tiles are [row][k] padded +8 elements: the row stride in dwords
 // is 4*odd, so the 16 lanes of a b128 fragment-read group (consecutive
 // rows, same k column) land on 16 distinct 4-dword bank slots.
-template <int BK, bool DBUF = true>
+template <int BK>
 struct FgemmShared {
   // BK=64 stages via DirectToLds (global_load_lds 16B): lane-packed
   // rows force the PACKED stride, with bank spread done by an in-row
   // granule XOR instead of padding.  BK=32 keeps the padded layout.
-  // DBUF=false: single-buffered (65.5 KB -> TWO workgroups per CU;
-  // load/compute overlap comes from the co-resident workgroup instead
-  // of the second buffer).
   static constexpr int SK = BK == 64 ? 64 : BK + 8;
-  static constexpr int NB = DBUF ? 2 : 1;
-  short a_tile[NB][kBM][SK];  // X tile, [m][k]
-  short b_tile[NB][kBN][SK];  // W tile, [n][k]
+  short a_tile[2][kBM][SK];  // X tile, [m][k]
+  short b_tile[2][kBN][SK];  // W tile, [n][k]
 };
 
 // XOR swizzle of the LDS column (in elements): the 16 lanes of a b128
@@ -84,8 +80,8 @@ __device__ __forceinline__ bf16x8 lds_read_b128(unsigned byte_addr) {
 // hipblaslt-ext GELU_AUX_BIAS epilogue this keeps the reference's erf
 // GELU numerics; ~10 extra VALU ops/element disappear under the MFMA
 // phases of a compute-bound tile)
-template <int EPI, int BK, bool DBUF = true>
-__global__ __launch_bounds__(kThreads, (BK == 64 && DBUF) ? 1 : 2) void fgemm_abt_kernel(
+template <int EPI, int BK>
+__global__ __launch_bounds__(kThreads, BK == 64 ? 1 : 2) void fgemm_abt_kernel(
     const short* __restrict__ a,     // [M, K]
     const short* __restrict__ b,     // [N, K]
     const short* __restrict__ bias,  // optional [N], may be null
@@ -93,11 +89,10 @@ __global__ __launch_bounds__(kThreads, (BK == 64 && DBUF) ? 1 : 2) void fgemm_ab
     short* __restrict__ aux,         // EPI==2: pre-GELU [M, N]
     int M, int N, int K) {
   constexpr bool WITH_BIAS = EPI >= 1;
-  constexpr int kSK = FgemmShared<BK, DBUF>::SK;
+  constexpr int kSK = FgemmShared<BK>::SK;
   constexpr int kChunks = BK / 32;
   HIP_DYNAMIC_SHARED(char, smem_raw)
-  FgemmShared<BK, DBUF>& sm =
-      *reinterpret_cast<FgemmShared<BK, DBUF>*>(smem_raw);
+  FgemmShared<BK>& sm = *reinterpret_cast<FgemmShared<BK>*>(smem_raw);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -275,22 +270,13 @@ __global__ __launch_bounds__(kThreads, (BK == 64 && DBUF) ? 1 : 2) void fgemm_ab
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    if (BK == 64 && DBUF) {
+    if (BK == 64) {
       if (ks + 1 < n_ksteps)
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __syncthreads();
-    } else if (BK == 64) {
-      // single buffer: reads done -> overwrite in place -> visible
-      if (ks + 1 < n_ksteps) {
-        __syncthreads();
-        issue_dtl(0, (long)(ks + 1) * BK);
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      __syncthreads();
-    } else {
-      if (ks + 1 < n_ksteps) write_tiles(buf ^ 1);
-      __syncthreads();
+    } else if (ks + 1 < n_ksteps) {
+      write_tiles(buf ^ 1);
     }
+    __syncthreads();
   }
 
   // epilogue: this lane owns rows m = ... + seg*4 + r of column n
@@ -485,8 +471,6 @@ template __global__ void fgemm_abt_kernel<2, 64>(const short*, const short*,
 template __global__ void fgemm_abt_kernel<0, 32>(const short*, const short*,
                                                  const short*, short*, short*,
                                                  int, int, int);
-template __global__ void fgemm_abt_kernel<0, 64, false>(
-    const short*, const short*, const short*, short*, short*, int, int, int);
 template __global__ void fgemm_abt_mi32_kernel<0>(const short*, const short*,
                                                   const short*, short*,
                                                   short*, int, int, int);
@@ -519,30 +503,20 @@ int fgemm_mi() {
   return mi;
 }
 
-template <int EPI, int BK, bool DBUF = true>
+template <int EPI, int BK>
 void launch_fgemm(const short* x, const short* w, const short* bias, short* c,
                   short* aux, long M, long N, long K, hipStream_t stream) {
   static bool attr_set = [] {
     (void)hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&fgemm_abt_kernel<EPI, BK, DBUF>),
-        hipFuncAttributeMaxDynamicSharedMemorySize,
-        sizeof(FgemmShared<BK, DBUF>));
+        reinterpret_cast<const void*>(&fgemm_abt_kernel<EPI, BK>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, sizeof(FgemmShared<BK>));
     return true;
   }();
   (void)attr_set;
   dim3 grid((unsigned)(M / kBM), (unsigned)(N / kBN));
-  hipLaunchKernelGGL((fgemm_abt_kernel<EPI, BK, DBUF>), grid, dim3(kThreads),
-                     sizeof(FgemmShared<BK, DBUF>), stream, x, w, bias, c, aux,
+  hipLaunchKernelGGL((fgemm_abt_kernel<EPI, BK>), grid, dim3(kThreads),
+                     sizeof(FgemmShared<BK>), stream, x, w, bias, c, aux,
                      (int)M, (int)N, (int)K);
-}
-
-// VITFSDP_FGEMM_SBUF=1: single-buffered LDS (2 workgroups/CU) A/B knob
-bool fgemm_sbuf() {
-  static bool v = [] {
-    const char* e = std::getenv("VITFSDP_FGEMM_SBUF");
-    return e != nullptr && std::atoi(e) == 1;
-  }();
-  return v;
 }
 
 template <int EPI>
@@ -570,8 +544,6 @@ void launch_fgemm_bk(const short* x, const short* w, const short* bias,
     launch_fgemm<EPI, 32>(x, w, bias, c, aux, M, N, K, stream);
   else if (fgemm_mi() == 32)
     launch_fgemm_mi32<EPI>(x, w, bias, c, aux, M, N, K, stream);
-  else if (fgemm_sbuf())
-    launch_fgemm<EPI, 64, false>(x, w, bias, c, aux, M, N, K, stream);
   else
     launch_fgemm<EPI, 64>(x, w, bias, c, aux, M, N, K, stream);
 }
