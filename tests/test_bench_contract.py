@@ -123,3 +123,26 @@ def test_auto_ckpt_blocks_formula():
     class Off(A):
         grad_ckpt = False
     assert ckpt(1, args=Off()) == -1
+
+
+def test_bench_torchrun_ws4():
+    """ws=4 on CPU/gloo: the driver's SCALE shape at a world size the
+    other tests don't cover (rank slicing, MAX-timing reduce, global
+    batch arithmetic, one JSON line)."""
+    port = _free_port()
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         "bench.py", "--gpus", "4", "--model", "vit-tiny",
+         "--per_gpu_batch", "2", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, cwd=REPO, timeout=900,
+    )
+    assert res.returncode == 0, res.stderr[-3000:]
+    json_lines = [ln for ln in res.stdout.splitlines()
+                  if ln.startswith("{") and ln.endswith("}")]
+    assert len(json_lines) == 1, res.stdout[-2000:]
+    out = json.loads(json_lines[0])
+    assert out["n_gpus"] == 4
+    assert out["config"]["global_batch"] == 8
+    assert out["config"]["parallelism"] == "fsdp4"
