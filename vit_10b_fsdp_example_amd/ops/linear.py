@@ -253,8 +253,8 @@ _FGEMM_MIN_TILES = int(os.environ.get("VITFSDP_FGEMM_MIN_TILES", "256"))
 
 def configure_gelu_fusion(embed_dim, hidden_dim):
     """Register the MLP dims so the dispatch mode can recognize fc1's
-    forward and fc2's dgrad by shape.  No-op when embed == hidden (the
-    shape keys would be ambiguous)."""
+    forward addmm by shape.  No-op when embed == hidden (the shape key
+    would be ambiguous)."""
     if embed_dim != hidden_dim:
         _GELU_CFG["d"] = int(embed_dim)
         _GELU_CFG["hid"] = int(hidden_dim)
