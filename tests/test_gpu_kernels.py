@@ -607,16 +607,24 @@ def test_fwd_gemm_gelu_erf_vs_reference():
     assert float((out.float() - out_ref).abs().max()) < 0.05
 
 
-def test_fmha_dropout_in_kernel_fwd_bwd():
+@pytest.mark.parametrize(
+    "B,H,T,D",
+    [
+        (2, 3, 128, 160),
+        (1, 1, 1024, 160),  # long context (448px class)
+        (1, 2, 200, 160),   # ragged: T not a tile multiple
+    ],
+)
+def test_fmha_dropout_in_kernel_fwd_bwd(B, H, T, D):
     """In-kernel attention dropout: the kernel must match the math
     composition evaluated with the EXACT mask (bit-reproduced from the
-    counter hash), forward and backward."""
+    counter hash), forward and backward — including long-context and
+    ragged tile shapes."""
     from vit_10b_fsdp_example_amd.ops.attention import (
         dropout_mask_reference,
     )
 
     torch.manual_seed(0)
-    B, H, T, D = 2, 3, 128, 160
     p, seed, scale = 0.3, 98765, D ** -0.5
     mk = lambda: (torch.randn(B, H, T, D, device=_dev()) * 0.5).to(
         torch.bfloat16
@@ -650,7 +658,7 @@ def test_fmha_dropout_in_kernel_fwd_bwd():
     # measured drop rate ~ p (the output of a row with all-ones V would
     # be the masked row-mean; check via probs-mask stats instead)
     keep_frac = mask.mean().item()
-    assert abs(keep_frac - (1 - p)) < 0.01
+    assert abs(keep_frac - (1 - p)) < 0.02
 
     # backward vs the exact-mask reference
     do = torch.randn_like(o)
